@@ -1,0 +1,3 @@
+"""oracle package — TEST INFRASTRUCTURE ONLY (see sptag_oracle.h).
+Only tests/, __graft_entry__.smoke() and bench.py's cpu_baseline leg may
+import this."""

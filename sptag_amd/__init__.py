@@ -1,0 +1,210 @@
+"""sptag_amd — MI355X-native SPTAG search backend (Python host mirror).
+
+This module mirrors the reference Python wrapper's ``AnnIndex`` facade
+(reference Wrappers/inc/CoreInterface.h:14: Load:72, Search:45,
+BatchSearch:49) over the C-ABI in include/sptag_amd.h. Host logic lives in
+the C++ library (sptag_amd/libsptag_amd.so); this wrapper only marshals
+numpy arrays across the C boundary.
+
+THE GPU IS THE PRODUCT PATH: importing works anywhere (so CPU-side tests can
+check symbol exports), but every search raises loudly when the HIP device or
+the extension is missing. There is no CPU fallback in this package.
+"""
+import ctypes
+import os
+
+import numpy as np
+
+__all__ = ["AnnIndex", "load_library", "gpu_available", "SptagAmdError"]
+
+_LIB = None
+_LIBPATH = os.path.join(os.path.dirname(os.path.abspath(__file__)), "libsptag_amd.so")
+
+VT_FLOAT, VT_INT8 = 0, 1
+DM_L2, DM_COSINE = 0, 1
+
+_ERRNAMES = {
+    0: "OK", -1: "IO", -2: "PARAM", -3: "NOGPU", -4: "UNSUPPORTED",
+    -5: "OOM", -6: "INTERNAL",
+}
+
+
+class SptagAmdError(RuntimeError):
+    def __init__(self, code, what=""):
+        super().__init__(f"sptag_amd error {_ERRNAMES.get(code, code)} {what}")
+        self.code = code
+
+
+def load_library():
+    """Load libsptag_amd.so; raises if the extension was not built."""
+    global _LIB
+    if _LIB is not None:
+        return _LIB
+    if not os.path.exists(_LIBPATH):
+        raise SptagAmdError(
+            -6, f"HIP extension missing: {_LIBPATH} — build it with "
+                "`make -C sptag_amd/csrc` (or __graft_entry__.build()); "
+                "this package has no CPU fallback")
+    lib = ctypes.CDLL(_LIBPATH)
+    lib.sptag_amd_load_index.restype = ctypes.c_void_p
+    lib.sptag_amd_load_index.argtypes = [ctypes.c_char_p, ctypes.c_int]
+    lib.sptag_amd_create_index.restype = ctypes.c_void_p
+    lib.sptag_amd_create_index.argtypes = [
+        ctypes.c_int32, ctypes.c_int32, ctypes.c_int, ctypes.c_int,
+        ctypes.c_void_p, ctypes.c_int32, ctypes.c_void_p, ctypes.c_int32,
+        ctypes.c_void_p, ctypes.c_int32, ctypes.c_void_p, ctypes.c_void_p,
+        ctypes.c_int]
+    lib.sptag_amd_free_index.argtypes = [ctypes.c_void_p]
+    lib.sptag_amd_search_batch.restype = ctypes.c_int
+    lib.sptag_amd_search_batch.argtypes = [
+        ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int32, ctypes.c_int32,
+        ctypes.c_int32, ctypes.c_void_p, ctypes.c_void_p]
+    lib.sptag_amd_truth.restype = ctypes.c_int
+    lib.sptag_amd_truth.argtypes = lib.sptag_amd_search_batch.argtypes[:4] + [
+        ctypes.c_void_p, ctypes.c_void_p]
+    lib.sptag_amd_save_index.restype = ctypes.c_int
+    lib.sptag_amd_save_index.argtypes = [ctypes.c_void_p, ctypes.c_char_p]
+    for f in ["num_vectors", "dim", "valuetype", "distmethod", "degree",
+              "default_maxcheck"]:
+        fn = getattr(lib, "sptag_amd_" + f)
+        fn.restype = ctypes.c_int32
+        fn.argtypes = [ctypes.c_void_p]
+    lib.sptag_amd_gpu_available.restype = ctypes.c_int
+    lib.sptag_amd_build_info.restype = ctypes.c_char_p
+    _LIB = lib
+    return lib
+
+
+def gpu_available():
+    return bool(load_library().sptag_amd_gpu_available())
+
+
+def _np_dtype(vt):
+    return np.float32 if vt == VT_FLOAT else np.int8
+
+
+class AnnIndex:
+    """Mirror of the reference Python AnnIndex facade over the GPU backend.
+
+    Reference method -> here:
+      AnnIndex.Load(folder)            -> AnnIndex.Load(folder)
+      AnnIndex.Search(query, k)        -> .Search(query, k)
+      AnnIndex.BatchSearch(q, n, k, _) -> .BatchSearch(queries, k)
+      (SaveIndex)                      -> .Save(folder)
+    """
+
+    def __init__(self, handle, owns=True):
+        self._lib = load_library()
+        self._h = handle
+        self._owns = owns
+        if not handle:
+            raise SptagAmdError(-1, "index load/create failed (see stderr)")
+
+    # -- construction -------------------------------------------------
+    @classmethod
+    def Load(cls, folder, device=0):
+        lib = load_library()
+        h = lib.sptag_amd_load_index(str(folder).encode(), device)
+        return cls(h)
+
+    @classmethod
+    def FromArrays(cls, vectors, tree_start, tree_nodes, graph, distmethod,
+                   deleted=None, device=0):
+        """Assemble from raw numpy blobs (file layouts without headers)."""
+        lib = load_library()
+        vectors = np.ascontiguousarray(vectors)
+        vt = VT_FLOAT if vectors.dtype == np.float32 else VT_INT8
+        if vectors.dtype not in (np.float32, np.int8):
+            raise ValueError("dtype must be float32 or int8")
+        n, dim = vectors.shape
+        tree_start = np.ascontiguousarray(tree_start, dtype=np.int32)
+        tree_nodes = np.ascontiguousarray(tree_nodes, dtype=np.int32)
+        graph = np.ascontiguousarray(graph, dtype=np.int32)
+        dm = {"L2": DM_L2, "Cosine": DM_COSINE}.get(distmethod, distmethod)
+        delp = None
+        if deleted is not None:
+            deleted = np.ascontiguousarray(deleted, dtype=np.uint8)
+            delp = deleted.ctypes.data_as(ctypes.c_void_p)
+        h = lib.sptag_amd_create_index(
+            n, dim, vt, dm, vectors.ctypes.data_as(ctypes.c_void_p),
+            len(tree_start), tree_start.ctypes.data_as(ctypes.c_void_p),
+            tree_nodes.size // 3, tree_nodes.ctypes.data_as(ctypes.c_void_p),
+            graph.shape[1], graph.ctypes.data_as(ctypes.c_void_p),
+            delp, device)
+        return cls(h)
+
+    def __del__(self):
+        if getattr(self, "_owns", False) and getattr(self, "_h", None):
+            self._lib.sptag_amd_free_index(self._h)
+            self._h = None
+
+    # -- metadata -----------------------------------------------------
+    @property
+    def n(self):
+        return self._lib.sptag_amd_num_vectors(self._h)
+
+    @property
+    def dim(self):
+        return self._lib.sptag_amd_dim(self._h)
+
+    @property
+    def valuetype(self):
+        return self._lib.sptag_amd_valuetype(self._h)
+
+    @property
+    def distmethod(self):
+        return self._lib.sptag_amd_distmethod(self._h)
+
+    @property
+    def degree(self):
+        return self._lib.sptag_amd_degree(self._h)
+
+    @property
+    def default_maxcheck(self):
+        return self._lib.sptag_amd_default_maxcheck(self._h)
+
+    # -- search -------------------------------------------------------
+    def BatchSearch(self, queries, k, max_check=0):
+        """GPU batched search. Returns (vids int32 [nq,k], dists f32 [nq,k]),
+        ascending by (dist, vid), vid=-1 padding — the contract of the
+        reference batch overload VectorIndex.h:103."""
+        queries = np.ascontiguousarray(queries, dtype=_np_dtype(self.valuetype))
+        if queries.ndim == 1:
+            queries = queries[None, :]
+        nq = queries.shape[0]
+        assert queries.shape[1] == self.dim, (queries.shape, self.dim)
+        vids = np.empty((nq, k), dtype=np.int32)
+        dists = np.empty((nq, k), dtype=np.float32)
+        rc = self._lib.sptag_amd_search_batch(
+            self._h, queries.ctypes.data_as(ctypes.c_void_p), nq, k, max_check,
+            vids.ctypes.data_as(ctypes.c_void_p),
+            dists.ctypes.data_as(ctypes.c_void_p))
+        if rc != 0:
+            raise SptagAmdError(rc, "search_batch")
+        return vids, dists
+
+    def Search(self, query, k, max_check=0):
+        vids, dists = self.BatchSearch(query, k, max_check)
+        return vids[0], dists[0]
+
+    def Truth(self, queries, k):
+        """Exact brute-force top-k on the GPU."""
+        queries = np.ascontiguousarray(queries, dtype=_np_dtype(self.valuetype))
+        if queries.ndim == 1:
+            queries = queries[None, :]
+        nq = queries.shape[0]
+        vids = np.empty((nq, k), dtype=np.int32)
+        dists = np.empty((nq, k), dtype=np.float32)
+        rc = self._lib.sptag_amd_truth(
+            self._h, queries.ctypes.data_as(ctypes.c_void_p), nq, k,
+            vids.ctypes.data_as(ctypes.c_void_p),
+            dists.ctypes.data_as(ctypes.c_void_p))
+        if rc != 0:
+            raise SptagAmdError(rc, "truth")
+        return vids, dists
+
+    def Save(self, folder):
+        os.makedirs(folder, exist_ok=True)
+        rc = self._lib.sptag_amd_save_index(self._h, str(folder).encode())
+        if rc != 0:
+            raise SptagAmdError(rc, "save_index")

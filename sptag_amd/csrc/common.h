@@ -1,0 +1,75 @@
+/* Internal host<->device shared definitions for the sptag_amd backend.
+ * Not part of the public ABI (that is include/sptag_amd.h). */
+#pragma once
+#include <stdint.h>
+
+namespace sptag_amd {
+
+/* mirror of the public constants */
+enum { VT_FLOAT = 0, VT_INT8 = 1 };
+enum { DM_L2 = 0, DM_COSINE = 1 };
+
+/* default search parameters (reference BKT/ParameterDefinitionList.h:47-49) */
+enum { DEFAULT_MAXCHECK = 8192, INIT_PIVOTS = 50, OTHER_PIVOTS = 4 };
+
+/* hard caps of the v1 kernel (host validates before launch) */
+enum { MAX_DEG = 64, MAX_K = 64, MAX_DIM = 4096 };
+
+/* device-resident index description (POD, passed by value to kernels) */
+struct DevIndex {
+    const void* vectors;       /* n*dim row-major, element size by vt */
+    const int32_t* graph;      /* n*deg row-major adjacency */
+    const int32_t* tree_nodes; /* 3*int32 {centerid,childStart,childEnd} per node */
+    const int32_t* tree_start; /* ntrees roots */
+    const uint8_t* deleted;    /* may be null */
+    int32_t n, dim, deg, ntrees, n_tree_nodes;
+    int32_t has_deleted;
+};
+
+/* per-launch search configuration */
+struct SearchCfg {
+    int32_t nq, k, max_check;
+    int32_t init_pivots, other_pivots;
+    int32_t ng_cap, spt_cap, dpq_cap;  /* heap capacities (entries) */
+    int32_t vcap;                      /* visited table slots (pow2) */
+};
+
+/* per-launch buffers */
+struct SearchBufs {
+    const void* queries;   /* nq*dim */
+    int32_t* out_vids;     /* nq*k */
+    float* out_dists;      /* nq*k */
+    int32_t* visited;      /* nq*vcap, zeroed before launch */
+    int32_t* oflow;        /* nq flags: nonzero => rerun with bigger caps */
+    /* only used by the global-heap variant: per-query strided scratch */
+    void* gheap_ng;        /* nq * (ng_cap+1) * 8B */
+    void* gheap_spt;       /* nq * (spt_cap+1) * 8B */
+};
+
+/* LDS bytes needed per workgroup for the LDS-heap variant (and the
+ * non-heap part of the global variant). Keep in sync with kernel. */
+inline size_t lds_bytes(int dim, size_t esz, const SearchCfg& c, bool heaps_in_lds)
+{
+    size_t b = 0;
+    b += ((size_t)dim * esz + 15) & ~15ul;            /* query vector */
+    b += (MAX_DEG) * 4;                               /* staged dists */
+    b += (MAX_DEG) * 4;                               /* staged child centers */
+    b += ((size_t)c.k) * 8;                           /* result set */
+    b += ((size_t)c.dpq_cap + 1) * 4;                 /* DistPriorityQueue */
+    b += 64;                                          /* scalar slots, padding */
+    if (heaps_in_lds) {
+        b += ((size_t)c.ng_cap + 1) * 8;
+        b += ((size_t)c.spt_cap + 1) * 8;
+    }
+    return b;
+}
+
+/* launchers implemented in kernels.hip; return hipError_t as int */
+int launch_bkt_search(int valuetype, int distmethod, bool heaps_in_lds,
+                      const DevIndex& di, const SearchCfg& cfg,
+                      const SearchBufs& bufs, void* stream);
+int launch_truth(int valuetype, int distmethod, const DevIndex& di,
+                 const void* queries, int32_t nq, int32_t k,
+                 int32_t* out_vids, float* out_dists, void* stream);
+
+}  /* namespace sptag_amd */

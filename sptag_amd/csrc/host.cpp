@@ -1,0 +1,558 @@
+/* sptag_amd host library — implements the C-ABI in include/sptag_amd.h.
+ *
+ * Host code is C++ (as the reference's is); all compute runs in the HIP
+ * kernels of kernels.hip. There is NO CPU search path here: if the HIP
+ * runtime or device is unavailable, every search call fails loudly with
+ * SPTAG_AMD_ERR_NOGPU.
+ *
+ * File formats are byte-compatible with the reference
+ * (/root/reference/AnnService):
+ *   vectors.bin  [int32 R][int32 C][row-major T]      Dataset.h:146
+ *   tree.bin     [int32 #trees][int32 roots x #]
+ *                [int32 count][{centerid,childStart,childEnd} x count]
+ *                                                     BKTree.h:640-686
+ *   graph.bin    [int32 R][int32 deg][int32 adj RxD]  NeighborhoodGraph.h:607
+ *   deletes.bin  [int32 count][int32 R][int32 1][int8 x R]  Labelset.h:78
+ *   indexloader.ini                                   VectorIndex.cpp:198-222
+ */
+#include "../../include/sptag_amd.h"
+
+#include <hip/hip_runtime.h>
+
+#include <cstdio>
+#include <cstring>
+#include <mutex>
+#include <string>
+#include <vector>
+
+#include "common.h"
+
+using namespace sptag_amd;
+
+namespace {
+
+struct HostIndex {
+    int device = 0;
+    int vt = VT_FLOAT, dm = DM_L2;
+    int32_t n = 0, dim = 0, deg = 0, ntrees = 0, n_tree_nodes = 0;
+    int32_t default_maxcheck = DEFAULT_MAXCHECK;
+    int32_t init_pivots = INIT_PIVOTS, other_pivots = OTHER_PIVOTS;
+    bool has_deleted = false;
+    /* device allocations */
+    void* d_vectors = nullptr;
+    int32_t* d_graph = nullptr;
+    int32_t* d_tree = nullptr;
+    int32_t* d_tree_start = nullptr;
+    uint8_t* d_deleted = nullptr;
+    /* host copies kept for save_index and shard assembly */
+    std::vector<char> h_vectors;
+    std::vector<int32_t> h_graph, h_tree, h_tree_start;
+    std::vector<uint8_t> h_deleted;
+    int64_t deleted_count = 0;
+    std::mutex lock;
+
+    size_t esz() const { return vt == VT_FLOAT ? 4 : 1; }
+    DevIndex dev() const {
+        DevIndex di;
+        di.vectors = d_vectors;
+        di.graph = d_graph;
+        di.tree_nodes = d_tree;
+        di.tree_start = d_tree_start;
+        di.deleted = d_deleted;
+        di.n = n; di.dim = dim; di.deg = deg; di.ntrees = ntrees;
+        di.n_tree_nodes = n_tree_nodes;
+        di.has_deleted = has_deleted ? 1 : 0;
+        return di;
+    }
+};
+
+#define HIP_OR_FAIL(expr, ret)                                              \
+    do {                                                                    \
+        hipError_t _e = (expr);                                             \
+        if (_e != hipSuccess) {                                             \
+            fprintf(stderr, "sptag_amd: HIP error %s at %s:%d\n",           \
+                    hipGetErrorString(_e), __FILE__, __LINE__);             \
+            return ret;                                                     \
+        }                                                                   \
+    } while (0)
+
+bool read_file(const std::string& path, std::vector<char>& out)
+{
+    FILE* f = fopen(path.c_str(), "rb");
+    if (!f) return false;
+    fseek(f, 0, SEEK_END);
+    long sz = ftell(f);
+    fseek(f, 0, SEEK_SET);
+    out.resize((size_t)sz);
+    bool ok = fread(out.data(), 1, (size_t)sz, f) == (size_t)sz;
+    fclose(f);
+    return ok;
+}
+
+bool ini_get(const std::string& text, const char* section, const char* key,
+             std::string& out)
+{
+    std::string sect = std::string("[") + section + "]";
+    size_t p = text.find(sect);
+    if (p == std::string::npos) return false;
+    p += sect.size();
+    size_t end = text.find('[', p);
+    std::string k = std::string(key) + "=";
+    while (p < text.size() && (end == std::string::npos || p < end)) {
+        size_t eol = text.find('\n', p);
+        if (eol == std::string::npos) eol = text.size();
+        while (p < eol && (text[p] == ' ' || text[p] == '\r')) p++;
+        if (text.compare(p, k.size(), k) == 0) {
+            out = text.substr(p + k.size(), eol - p - k.size());
+            while (!out.empty() && (out.back() == '\r' || out.back() == ' '))
+                out.pop_back();
+            return true;
+        }
+        p = eol + 1;
+    }
+    return false;
+}
+
+int upload(HostIndex* ix)
+{
+    HIP_OR_FAIL(hipSetDevice(ix->device), SPTAG_AMD_ERR_NOGPU);
+    HIP_OR_FAIL(hipMalloc(&ix->d_vectors, ix->h_vectors.size()), SPTAG_AMD_ERR_OOM);
+    HIP_OR_FAIL(hipMemcpy(ix->d_vectors, ix->h_vectors.data(), ix->h_vectors.size(),
+                          hipMemcpyHostToDevice), SPTAG_AMD_ERR_NOGPU);
+    HIP_OR_FAIL(hipMalloc(&ix->d_graph, ix->h_graph.size() * 4), SPTAG_AMD_ERR_OOM);
+    HIP_OR_FAIL(hipMemcpy(ix->d_graph, ix->h_graph.data(), ix->h_graph.size() * 4,
+                          hipMemcpyHostToDevice), SPTAG_AMD_ERR_NOGPU);
+    HIP_OR_FAIL(hipMalloc(&ix->d_tree, ix->h_tree.size() * 4), SPTAG_AMD_ERR_OOM);
+    HIP_OR_FAIL(hipMemcpy(ix->d_tree, ix->h_tree.data(), ix->h_tree.size() * 4,
+                          hipMemcpyHostToDevice), SPTAG_AMD_ERR_NOGPU);
+    HIP_OR_FAIL(hipMalloc(&ix->d_tree_start, ix->h_tree_start.size() * 4), SPTAG_AMD_ERR_OOM);
+    HIP_OR_FAIL(hipMemcpy(ix->d_tree_start, ix->h_tree_start.data(),
+                          ix->h_tree_start.size() * 4, hipMemcpyHostToDevice),
+                SPTAG_AMD_ERR_NOGPU);
+    if (ix->has_deleted) {
+        HIP_OR_FAIL(hipMalloc(&ix->d_deleted, ix->h_deleted.size()), SPTAG_AMD_ERR_OOM);
+        HIP_OR_FAIL(hipMemcpy(ix->d_deleted, ix->h_deleted.data(), ix->h_deleted.size(),
+                              hipMemcpyHostToDevice), SPTAG_AMD_ERR_NOGPU);
+    }
+    return SPTAG_AMD_OK;
+}
+
+uint32_t next_pow2(uint32_t x)
+{
+    uint32_t p = 1;
+    while (p < x) p <<= 1;
+    return p;
+}
+
+}  // namespace
+
+struct SptagAmdIndex : HostIndex {};
+
+extern "C" {
+
+int sptag_amd_gpu_available(void)
+{
+    int n = 0;
+    return hipGetDeviceCount(&n) == hipSuccess && n > 0;
+}
+
+const char* sptag_amd_build_info(void)
+{
+    return "sptag_amd gfx950 v1 (BKT search; HIP " __DATE__ ")";
+}
+
+SptagAmdIndex* sptag_amd_create_index(int32_t n, int32_t dim, int valuetype,
+                                      int distmethod, const void* vectors,
+                                      int32_t ntrees, const int32_t* tree_start,
+                                      int32_t n_tree_nodes, const int32_t* tree_nodes,
+                                      int32_t degree, const int32_t* graph,
+                                      const uint8_t* deleted, int device)
+{
+    if (n <= 0 || dim <= 0 || dim > MAX_DIM || degree <= 0 || degree > MAX_DEG ||
+        ntrees <= 0 || n_tree_nodes <= 0)
+        return nullptr;
+    if (valuetype == SPTAG_AMD_VT_INT8 && (int64_t)dim * 254 * 254 >= (1ll << 24)) {
+        fprintf(stderr, "sptag_amd: int8 dim %d exceeds exact-float range\n", dim);
+        return nullptr;
+    }
+    auto* ix = new SptagAmdIndex();
+    ix->device = device;
+    ix->vt = valuetype;
+    ix->dm = distmethod;
+    ix->n = n; ix->dim = dim; ix->deg = degree;
+    ix->ntrees = ntrees;
+    ix->h_vectors.assign((const char*)vectors,
+                         (const char*)vectors + (size_t)n * dim * ix->esz());
+    ix->h_graph.assign(graph, graph + (size_t)n * degree);
+    ix->h_tree_start.assign(tree_start, tree_start + ntrees);
+    ix->h_tree.assign(tree_nodes, tree_nodes + (size_t)n_tree_nodes * 3);
+    /* BKTree.h:683: append sentinel if the stored array does not end with
+     * centerid == -1 */
+    if (ix->h_tree[(size_t)(n_tree_nodes - 1) * 3] != -1) {
+        ix->h_tree.push_back(-1); ix->h_tree.push_back(-1); ix->h_tree.push_back(-1);
+        n_tree_nodes += 1;
+    }
+    ix->n_tree_nodes = n_tree_nodes;
+    if (deleted) {
+        ix->h_deleted.assign(deleted, deleted + n);
+        for (int32_t i = 0; i < n; i++) ix->deleted_count += deleted[i] ? 1 : 0;
+        ix->has_deleted = ix->deleted_count > 0;
+    }
+    if (sptag_amd_gpu_available()) {
+        if (upload(ix) != SPTAG_AMD_OK) {
+            sptag_amd_free_index(ix);
+            return nullptr;
+        }
+    }
+    /* without a GPU the handle still supports metadata queries + save_index;
+     * search fails with ERR_NOGPU. */
+    return ix;
+}
+
+SptagAmdIndex* sptag_amd_load_index(const char* folder, int device)
+{
+    std::string dir(folder);
+    if (!dir.empty() && dir.back() != '/') dir += '/';
+
+    std::vector<char> raw;
+    if (!read_file(dir + "indexloader.ini", raw)) return nullptr;
+    std::string ini(raw.begin(), raw.end());
+
+    std::string val;
+    if (ini_get(ini, "Index", "IndexAlgoType", val) && val != "BKT") {
+        fprintf(stderr, "sptag_amd: IndexAlgoType %s not supported yet (BKT only)\n",
+                val.c_str());
+        return nullptr;
+    }
+    int vt = VT_FLOAT, dm = DM_L2;
+    if (ini_get(ini, "Index", "ValueType", val)) {
+        if (val == "Float") vt = VT_FLOAT;
+        else if (val == "Int8") vt = VT_INT8;
+        else {
+            fprintf(stderr, "sptag_amd: ValueType %s not supported\n", val.c_str());
+            return nullptr;
+        }
+    }
+    if (ini_get(ini, "Index", "DistCalcMethod", val)) {
+        if (val == "L2") dm = DM_L2;
+        else if (val == "Cosine" || val == "InnerProduct") dm = DM_COSINE;
+        else {
+            fprintf(stderr, "sptag_amd: DistCalcMethod %s not supported\n", val.c_str());
+            return nullptr;
+        }
+    }
+
+    std::vector<char> vb, tb, gb, db;
+    if (!read_file(dir + "vectors.bin", vb) || !read_file(dir + "tree.bin", tb) ||
+        !read_file(dir + "graph.bin", gb))
+        return nullptr;
+    bool have_del = read_file(dir + "deletes.bin", db);
+
+    const int32_t* vh = (const int32_t*)vb.data();
+    int32_t n = vh[0], dim = vh[1];
+    const int32_t* th = (const int32_t*)tb.data();
+    int32_t ntrees = th[0];
+    const int32_t* tstart = th + 1;
+    int32_t nnodes = th[1 + ntrees];
+    const int32_t* tnodes = th + 2 + ntrees;
+    const int32_t* gh = (const int32_t*)gb.data();
+    int32_t gn = gh[0], deg = gh[1];
+    if (gn != n) {
+        fprintf(stderr, "sptag_amd: graph rows %d != vectors %d\n", gn, n);
+        return nullptr;
+    }
+    const uint8_t* del = nullptr;
+    if (have_del && db.size() >= 12 + (size_t)n) {
+        int32_t delcount = ((const int32_t*)db.data())[0];
+        if (delcount > 0) del = (const uint8_t*)db.data() + 12;
+    }
+
+    auto* ix = sptag_amd_create_index(n, dim, vt, dm, vb.data() + 8, ntrees, tstart,
+                                      nnodes, tnodes, deg, gh + 2, del, device);
+    if (!ix) return nullptr;
+    if (ini_get(ini, "Index", "MaxCheck", val)) ix->default_maxcheck = atoi(val.c_str());
+    if (ini_get(ini, "Index", "NumberOfInitialDynamicPivots", val))
+        ix->init_pivots = atoi(val.c_str());
+    if (ini_get(ini, "Index", "NumberOfOtherDynamicPivots", val))
+        ix->other_pivots = atoi(val.c_str());
+    if (ini_get(ini, "Index", "EnableBfs", val) && atoi(val.c_str()) != 0) {
+        fprintf(stderr, "sptag_amd: EnableBfs not supported\n");
+        sptag_amd_free_index(ix);
+        return nullptr;
+    }
+    return ix;
+}
+
+void sptag_amd_free_index(SptagAmdIndex* ix)
+{
+    if (!ix) return;
+    if (ix->d_vectors) hipFree(ix->d_vectors);
+    if (ix->d_graph) hipFree(ix->d_graph);
+    if (ix->d_tree) hipFree(ix->d_tree);
+    if (ix->d_tree_start) hipFree(ix->d_tree_start);
+    if (ix->d_deleted) hipFree(ix->d_deleted);
+    delete ix;
+}
+
+int sptag_amd_search_batch(SptagAmdIndex* ix, const void* queries, int32_t nq,
+                           int32_t k, int32_t max_check,
+                           int32_t* out_vids, float* out_dists)
+{
+    if (!ix || !queries || nq <= 0 || k <= 0 || k > MAX_K) return SPTAG_AMD_ERR_PARAM;
+    if (!sptag_amd_gpu_available() || !ix->d_vectors) {
+        fprintf(stderr,
+                "sptag_amd: search requires a HIP device (%s); no CPU fallback\n",
+                sptag_amd_build_info());
+        return SPTAG_AMD_ERR_NOGPU;
+    }
+    std::lock_guard<std::mutex> g(ix->lock);
+    HIP_OR_FAIL(hipSetDevice(ix->device), SPTAG_AMD_ERR_NOGPU);
+
+    if (max_check <= 0) max_check = ix->default_maxcheck;
+
+    SearchCfg cfg;
+    cfg.nq = nq;
+    cfg.k = k;
+    cfg.max_check = max_check;
+    cfg.init_pivots = ix->init_pivots;
+    cfg.other_pivots = ix->other_pivots;
+    cfg.dpq_cap = std::max(max_check / 16, k);   /* WorkSpace.h:268 */
+    cfg.vcap = (int32_t)next_pow2((uint32_t)std::max(4096, max_check * 6));
+    /* LDS-variant capacities: generous slack over the expected occupancy
+     * (~checked + tree inserts); overflow reruns use reference capacities. */
+    cfg.ng_cap = max_check + 4096;
+    cfg.spt_cap = 2048;
+
+    /* dynamic-LDS ceiling differs from the 160 KiB hardware LDS on some
+     * runtimes; ask the device and fall back to global heaps when the
+     * budget does not fit. */
+    int lds_limit = 64 * 1024;
+    (void)hipDeviceGetAttribute(&lds_limit, hipDeviceAttributeMaxSharedMemoryPerBlock,
+                                ix->device);
+    bool lds_ok = lds_bytes(ix->dim, ix->esz(), cfg, true) <= (size_t)lds_limit;
+
+    /* device buffers */
+    size_t qbytes = (size_t)nq * ix->dim * ix->esz();
+    void* d_q = nullptr;
+    int32_t* d_vids = nullptr;
+    float* d_dists = nullptr;
+    int32_t* d_visited = nullptr;
+    int32_t* d_oflow = nullptr;
+    void* d_gng = nullptr;
+    void* d_gspt = nullptr;
+    int rc = SPTAG_AMD_ERR_NOGPU;
+    std::vector<int32_t> oflow(nq);
+
+    do {
+        if (hipMalloc(&d_q, qbytes) != hipSuccess) { rc = SPTAG_AMD_ERR_OOM; break; }
+        if (hipMemcpy(d_q, queries, qbytes, hipMemcpyHostToDevice) != hipSuccess) break;
+        if (hipMalloc(&d_vids, (size_t)nq * k * 4) != hipSuccess) { rc = SPTAG_AMD_ERR_OOM; break; }
+        if (hipMalloc(&d_dists, (size_t)nq * k * 4) != hipSuccess) { rc = SPTAG_AMD_ERR_OOM; break; }
+        if (hipMalloc(&d_oflow, (size_t)nq * 4) != hipSuccess) { rc = SPTAG_AMD_ERR_OOM; break; }
+        if (hipMalloc(&d_visited, (size_t)nq * cfg.vcap * 4) != hipSuccess) { rc = SPTAG_AMD_ERR_OOM; break; }
+        if (hipMemset(d_visited, 0, (size_t)nq * cfg.vcap * 4) != hipSuccess) break;
+
+        SearchBufs bufs;
+        bufs.queries = d_q;
+        bufs.out_vids = d_vids;
+        bufs.out_dists = d_dists;
+        bufs.visited = d_visited;
+        bufs.oflow = d_oflow;
+        bufs.gheap_ng = nullptr;
+        bufs.gheap_spt = nullptr;
+
+        if (lds_ok) {
+            int err = launch_bkt_search(ix->vt, ix->dm, true, ix->dev(), cfg, bufs, nullptr);
+            if (err != 0) { fprintf(stderr, "sptag_amd: launch failed %d\n", err); break; }
+            if (hipDeviceSynchronize() != hipSuccess) break;
+            if (hipMemcpy(oflow.data(), d_oflow, (size_t)nq * 4, hipMemcpyDeviceToHost)
+                != hipSuccess) break;
+        } else {
+            for (auto& f : oflow) f = 1;   /* go straight to the global variant */
+        }
+
+        /* rerun overflowed queries with the global-heap variant at the
+         * reference's own capacities (WorkSpace.h:265). */
+        std::vector<int32_t> redo;
+        for (int32_t i = 0; i < nq; i++)
+            if (oflow[i]) redo.push_back(i);
+        if (!redo.empty()) {
+            SearchCfg c2 = cfg;
+            c2.nq = (int32_t)redo.size();
+            c2.ng_cap = max_check * 30;
+            c2.spt_cap = max_check * 10;
+            std::vector<char> q2((size_t)redo.size() * ix->dim * ix->esz());
+            for (size_t i = 0; i < redo.size(); i++)
+                memcpy(q2.data() + i * ix->dim * ix->esz(),
+                       (const char*)queries + (size_t)redo[i] * ix->dim * ix->esz(),
+                       ix->dim * ix->esz());
+            void* d_q2 = nullptr;
+            int32_t* d_v2 = nullptr;
+            float* d_d2 = nullptr;
+            if (hipMalloc(&d_q2, q2.size()) != hipSuccess ||
+                hipMalloc(&d_v2, redo.size() * k * 4) != hipSuccess ||
+                hipMalloc(&d_d2, redo.size() * k * 4) != hipSuccess ||
+                hipMalloc(&d_gng, redo.size() * ((size_t)c2.ng_cap + 1) * 8) != hipSuccess ||
+                hipMalloc(&d_gspt, redo.size() * ((size_t)c2.spt_cap + 1) * 8) != hipSuccess) {
+                rc = SPTAG_AMD_ERR_OOM;
+                if (d_q2) hipFree(d_q2);
+                if (d_v2) hipFree(d_v2);
+                if (d_d2) hipFree(d_d2);
+                break;
+            }
+            hipMemcpy(d_q2, q2.data(), q2.size(), hipMemcpyHostToDevice);
+            hipMemset(d_visited, 0, (size_t)redo.size() * cfg.vcap * 4);
+            hipMemset(d_oflow, 0, (size_t)redo.size() * 4);
+            SearchBufs b2 = bufs;
+            b2.queries = d_q2;
+            b2.out_vids = d_v2;
+            b2.out_dists = d_d2;
+            b2.gheap_ng = d_gng;
+            b2.gheap_spt = d_gspt;
+            int err = launch_bkt_search(ix->vt, ix->dm, false, ix->dev(), c2, b2, nullptr);
+            if (err != 0 || hipDeviceSynchronize() != hipSuccess) {
+                fprintf(stderr, "sptag_amd: fallback launch failed %d\n", err);
+                hipFree(d_q2); hipFree(d_v2); hipFree(d_d2);
+                break;
+            }
+            std::vector<int32_t> v2(redo.size() * k);
+            std::vector<float> dd2(redo.size() * k);
+            hipMemcpy(v2.data(), d_v2, v2.size() * 4, hipMemcpyDeviceToHost);
+            hipMemcpy(dd2.data(), d_d2, dd2.size() * 4, hipMemcpyDeviceToHost);
+            hipFree(d_q2); hipFree(d_v2); hipFree(d_d2);
+            /* scatter into the main output after the bulk copy below */
+            if (hipMemcpy(out_vids, d_vids, (size_t)nq * k * 4, hipMemcpyDeviceToHost)
+                != hipSuccess) break;
+            if (hipMemcpy(out_dists, d_dists, (size_t)nq * k * 4, hipMemcpyDeviceToHost)
+                != hipSuccess) break;
+            for (size_t i = 0; i < redo.size(); i++) {
+                memcpy(out_vids + (size_t)redo[i] * k, v2.data() + i * k, (size_t)k * 4);
+                memcpy(out_dists + (size_t)redo[i] * k, dd2.data() + i * k, (size_t)k * 4);
+            }
+            rc = SPTAG_AMD_OK;
+            break;
+        }
+
+        if (hipMemcpy(out_vids, d_vids, (size_t)nq * k * 4, hipMemcpyDeviceToHost)
+            != hipSuccess) break;
+        if (hipMemcpy(out_dists, d_dists, (size_t)nq * k * 4, hipMemcpyDeviceToHost)
+            != hipSuccess) break;
+        rc = SPTAG_AMD_OK;
+    } while (0);
+
+    if (d_q) hipFree(d_q);
+    if (d_vids) hipFree(d_vids);
+    if (d_dists) hipFree(d_dists);
+    if (d_visited) hipFree(d_visited);
+    if (d_oflow) hipFree(d_oflow);
+    if (d_gng) hipFree(d_gng);
+    if (d_gspt) hipFree(d_gspt);
+    return rc;
+}
+
+int sptag_amd_truth(SptagAmdIndex* ix, const void* queries, int32_t nq,
+                    int32_t k, int32_t* out_vids, float* out_dists)
+{
+    if (!ix || !queries || nq <= 0 || k <= 0 || k > MAX_K) return SPTAG_AMD_ERR_PARAM;
+    if (!sptag_amd_gpu_available() || !ix->d_vectors) return SPTAG_AMD_ERR_NOGPU;
+    std::lock_guard<std::mutex> g(ix->lock);
+    HIP_OR_FAIL(hipSetDevice(ix->device), SPTAG_AMD_ERR_NOGPU);
+
+    size_t qbytes = (size_t)nq * ix->dim * ix->esz();
+    void* d_q = nullptr;
+    int32_t* d_v = nullptr;
+    float* d_d = nullptr;
+    HIP_OR_FAIL(hipMalloc(&d_q, qbytes), SPTAG_AMD_ERR_OOM);
+    HIP_OR_FAIL(hipMalloc(&d_v, (size_t)nq * k * 4), SPTAG_AMD_ERR_OOM);
+    HIP_OR_FAIL(hipMalloc(&d_d, (size_t)nq * k * 4), SPTAG_AMD_ERR_OOM);
+    hipMemcpy(d_q, queries, qbytes, hipMemcpyHostToDevice);
+    int err = launch_truth(ix->vt, ix->dm, ix->dev(), d_q, nq, k, d_v, d_d, nullptr);
+    int rc = SPTAG_AMD_ERR_NOGPU;
+    if (err == 0 && hipDeviceSynchronize() == hipSuccess) {
+        hipMemcpy(out_vids, d_v, (size_t)nq * k * 4, hipMemcpyDeviceToHost);
+        hipMemcpy(out_dists, d_d, (size_t)nq * k * 4, hipMemcpyDeviceToHost);
+        rc = SPTAG_AMD_OK;
+    }
+    hipFree(d_q); hipFree(d_v); hipFree(d_d);
+    return rc;
+}
+
+int sptag_amd_save_index(SptagAmdIndex* ix, const char* folder)
+{
+    if (!ix || !folder) return SPTAG_AMD_ERR_PARAM;
+    std::string dir(folder);
+    if (!dir.empty() && dir.back() != '/') dir += '/';
+
+    auto wfile = [&](const std::string& name, const void* hdr, size_t hdrsz,
+                     const void* body, size_t bodysz) -> bool {
+        FILE* f = fopen((dir + name).c_str(), "wb");
+        if (!f) return false;
+        bool ok = fwrite(hdr, 1, hdrsz, f) == hdrsz &&
+                  (bodysz == 0 || fwrite(body, 1, bodysz, f) == bodysz);
+        fclose(f);
+        return ok;
+    };
+
+    int32_t vh[2] = {ix->n, ix->dim};
+    if (!wfile("vectors.bin", vh, 8, ix->h_vectors.data(), ix->h_vectors.size()))
+        return SPTAG_AMD_ERR_IO;
+    {
+        FILE* f = fopen((dir + "tree.bin").c_str(), "wb");
+        if (!f) return SPTAG_AMD_ERR_IO;
+        fwrite(&ix->ntrees, 4, 1, f);
+        fwrite(ix->h_tree_start.data(), 4, ix->ntrees, f);
+        fwrite(&ix->n_tree_nodes, 4, 1, f);
+        fwrite(ix->h_tree.data(), 12, ix->n_tree_nodes, f);
+        fclose(f);
+    }
+    int32_t gh[2] = {ix->n, ix->deg};
+    if (!wfile("graph.bin", gh, 8, ix->h_graph.data(), ix->h_graph.size() * 4))
+        return SPTAG_AMD_ERR_IO;
+    {
+        std::vector<uint8_t> del = ix->h_deleted;
+        if (del.empty()) del.assign((size_t)ix->n, 0);
+        int32_t dh[3] = {(int32_t)ix->deleted_count, ix->n, 1};
+        if (!wfile("deletes.bin", dh, 12, del.data(), del.size()))
+            return SPTAG_AMD_ERR_IO;
+    }
+    {
+        FILE* f = fopen((dir + "indexloader.ini").c_str(), "wb");
+        if (!f) return SPTAG_AMD_ERR_IO;
+        fprintf(f,
+                "[Index]\n"
+                "IndexAlgoType=BKT\n"
+                "ValueType=%s\n\n"
+                "TreeFilePath=tree.bin\nGraphFilePath=graph.bin\n"
+                "VectorFilePath=vectors.bin\nDeleteVectorFilePath=deletes.bin\n"
+                "EnableBfs=0\nBKTNumber=%d\nBKTKmeansK=32\nBKTLeafSize=8\n"
+                "Samples=1000\nBKTLambdaFactor=100.000000\nTPTNumber=32\n"
+                "TPTLeafSize=2000\nNumTopDimensionTpTreeSplit=5\n"
+                "NeighborhoodSize=%d\nGraphNeighborhoodScale=2.000000\n"
+                "GraphCEFScale=2.000000\nRefineIterations=2\nEnableRebuild=0\n"
+                "CEF=1000\nAddCEF=500\nMaxCheckForRefineGraph=8192\n"
+                "RNGFactor=1.000000\nGPUGraphType=2\nGPURefineSteps=0\n"
+                "GPURefineDepth=30\nGPULeafSize=500\nHeadNumGPUs=1\n"
+                "TPTBalanceFactor=2\nNumberOfThreads=4\nDistCalcMethod=%s\n"
+                "DeletePercentageForRefine=0.400000\nAddCountForRebuild=1000\n"
+                "MaxCheck=%d\n"
+                "ThresholdOfNumberOfContinuousNoBetterPropagation=3\n"
+                "NumberOfInitialDynamicPivots=%d\nNumberOfOtherDynamicPivots=%d\n"
+                "HashTableExponent=2\nDataBlockSize=1048576\n"
+                "DataCapacity=2147483647\nMetaRecordSize=10\n",
+                ix->vt == VT_FLOAT ? "Float" : "Int8", ix->ntrees, ix->deg,
+                ix->dm == DM_L2 ? "L2" : "Cosine", ix->default_maxcheck,
+                ix->init_pivots, ix->other_pivots);
+        fclose(f);
+    }
+    return SPTAG_AMD_OK;
+}
+
+int32_t sptag_amd_num_vectors(const SptagAmdIndex* ix) { return ix ? ix->n : -1; }
+int32_t sptag_amd_dim(const SptagAmdIndex* ix) { return ix ? ix->dim : -1; }
+int sptag_amd_valuetype(const SptagAmdIndex* ix) { return ix ? ix->vt : -1; }
+int sptag_amd_distmethod(const SptagAmdIndex* ix) { return ix ? ix->dm : -1; }
+int32_t sptag_amd_degree(const SptagAmdIndex* ix) { return ix ? ix->deg : -1; }
+int32_t sptag_amd_default_maxcheck(const SptagAmdIndex* ix)
+{ return ix ? ix->default_maxcheck : -1; }
+
+}  /* extern "C" */

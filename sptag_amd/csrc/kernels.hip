@@ -1,0 +1,685 @@
+/* sptag_amd GPU kernels — gfx950 (MI355X, CDNA4) native.
+ *
+ * Batched BKT best-first graph search: one query per 64-thread workgroup
+ * (one wave64). The traversal replicates the reference CPU algorithm
+ * decision-for-decision (citations to /root/reference/AnnService):
+ *   - seed phase: BKTree::InitSearchTrees / SearchTrees (BKTree.h:697,772)
+ *   - traversal:  BKT::Index<T>::Search<> loop (BKTIndex.cpp:272-352)
+ *   - frontier:   Heap<NodeDistPair> semantics (Heap.h:14-110)
+ *   - pruning:    DistPriorityQueue (WorkSpace.h:167-225)
+ *   - top-k:      QueryResultSet AddPoint/SortResult (QueryResultSet.h:31)
+ * Distances reproduce the reference binary's float summation order
+ * (AVX512 16-lane chunking with fused mul+add; exact integer math for int8),
+ * so int8 results are bit-exact and float results are bit-exact against
+ * the oracle/_ref build (see oracle/sptag_oracle.c).
+ *
+ * Division of labor inside the wave:
+ *   - heap/result-set mutation is inherently serial (pop order on equal
+ *     keys is part of the parity contract) -> lane 0, state in LDS;
+ *   - adjacency row load: lanes 0..deg-1, one coalesced 128B line;
+ *   - visited-set: per-query open-addressing table in GLOBAL memory
+ *     (512KB-class tables do not fit LDS at MaxCheck 8192 — SURVEY.md §7),
+ *     probed in parallel with atomicCAS (set semantics only, order-free);
+ *   - distances: 16-lane groups, 4 candidates in flight per wave.
+ *
+ * HEAPS_IN_LDS=true keeps the NG/SPT frontier heaps in LDS with reduced
+ * capacities; if a heap would exceed its reduced capacity while the
+ * reference's (30*maxCheck / 10*maxCheck, WorkSpace.h:265) would not, the
+ * query sets an overflow flag and the host reruns it with the global-memory
+ * variant at full reference capacities — never a silent semantic change.
+ */
+#include <hip/hip_runtime.h>
+
+#include "common.h"
+
+namespace sptag_amd {
+
+#define DEV __device__ __forceinline__
+
+static constexpr float MAXDIST = __FLT_MAX__ / 10.0f; /* Common.h:122 */
+
+struct NodeDist { int32_t node; float distance; };
+struct QRes { int32_t vid; float dist; };
+
+/* ------------------------------------------------------------------ *
+ * serial structures (lane 0 only), exact reference semantics
+ * ------------------------------------------------------------------ */
+
+/* 2^floor(log2(size)) — start of the heap's last level (Heap.h:25) */
+DEV int heap_lastlevel(int size) { return 1 << (31 - __clz(size)); }
+
+struct HeapRef {
+    NodeDist* a;   /* 1-based; a[0] = empty-top sentinel {-1, MAXDIST} */
+    int cap;       /* this buffer's capacity */
+    int ref_cap;   /* the reference's capacity for this heap */
+};
+
+/* Heap.h:38-62 insert (incl. the full-heap last-level replace path). */
+DEV void ndheap_insert(HeapRef h, int* count, NodeDist v, int* oflow)
+{
+    int loc;
+    if (*count == h.cap) {
+        if (h.cap < h.ref_cap) { *oflow = 1; return; }
+        int lastlevel = heap_lastlevel(h.cap);
+        int maxi = lastlevel;
+        for (int i = lastlevel + 1; i <= h.cap; i++)
+            if (h.a[maxi].distance < h.a[i].distance) maxi = i;
+        if (v.distance > h.a[maxi].distance) return;
+        loc = maxi;
+    } else {
+        loc = ++(*count);
+    }
+    int par = loc >> 1;
+    while (par > 0 && v.distance < h.a[par].distance) {
+        h.a[loc] = h.a[par];
+        loc = par;
+        par >>= 1;
+    }
+    h.a[loc] = v;
+}
+
+/* Heap.h:90-105 heapify + :74-82 pop */
+DEV void ndheap_heapify(HeapRef h, int count)
+{
+    int parent = 1, next = 2;
+    while (next < count) {
+        if (h.a[next].distance > h.a[next + 1].distance) next++;
+        if (h.a[next].distance < h.a[parent].distance) {
+            NodeDist t = h.a[parent]; h.a[parent] = h.a[next]; h.a[next] = t;
+            parent = next;
+            next <<= 1;
+        } else break;
+    }
+    if (next == count && h.a[next].distance < h.a[parent].distance) {
+        NodeDist t = h.a[parent]; h.a[parent] = h.a[next]; h.a[next] = t;
+    }
+}
+
+DEV NodeDist ndheap_pop(HeapRef h, int* count)
+{
+    if (*count == 0) return h.a[0];
+    NodeDist t = h.a[1]; h.a[1] = h.a[*count]; h.a[*count] = t;
+    (*count)--;
+    ndheap_heapify(h, *count);
+    return h.a[*count + 1];
+}
+
+DEV NodeDist ndheap_top(HeapRef h, int count) { return count == 0 ? h.a[0] : h.a[1]; }
+
+/* DistPriorityQueue (WorkSpace.h:167-225): 1-based bounded float max-heap
+ * seeded with MaxDist; len starts at 1. */
+DEV int dpq_insert(float* a, int* len, int cap, float dist)
+{
+    if (dist > a[1]) return 0;
+    if (*len == cap) {
+        a[1] = dist;
+        int parent = 1, next = 2;
+        while (next < *len) {
+            if (a[next] < a[next + 1]) next++;
+            if (a[next] > a[parent]) {
+                float t = a[parent]; a[parent] = a[next]; a[next] = t;
+                parent = next;
+                next <<= 1;
+            } else break;
+        }
+        if (next == *len && a[next] > a[parent]) {
+            float t = a[parent]; a[parent] = a[next]; a[next] = t;
+        }
+    } else {
+        int next = ++(*len), parent = next >> 1;
+        while (parent > 0 && dist > a[parent]) {
+            a[next] = a[parent];
+            next = parent;
+            parent >>= 1;
+        }
+        a[next] = dist;
+    }
+    return 1;
+}
+
+/* QueryResultSet (QueryResultSet.h:17-120): 0-based k-entry max-heap on
+ * (dist, vid). */
+DEV int qres_lt(QRes x, QRes y)
+{
+    return (x.dist < y.dist) || (x.dist == y.dist && x.vid < y.vid);
+}
+
+DEV void qrs_heapify(QRes* r, int count)
+{
+    int parent = 0, next = 1, maxidx = count - 1;
+    while (next < maxidx) {
+        if (qres_lt(r[next], r[next + 1])) next++;
+        if (qres_lt(r[parent], r[next])) {
+            QRes t = r[next]; r[next] = r[parent]; r[parent] = t;
+            parent = next;
+            next = (parent << 1) + 1;
+        } else break;
+    }
+    if (next == maxidx && qres_lt(r[parent], r[next])) {
+        QRes t = r[next]; r[next] = r[parent]; r[parent] = t;
+    }
+}
+
+DEV int qrs_add(QRes* r, int k, int32_t vid, float dist)
+{
+    if (dist < r[0].dist || (dist == r[0].dist && vid < r[0].vid)) {
+        r[0].vid = vid; r[0].dist = dist;
+        qrs_heapify(r, k);
+        return 1;
+    }
+    return 0;
+}
+
+DEV void qrs_sort(QRes* r, int k)
+{
+    for (int i = k - 1; i >= 0; i--) {
+        QRes t = r[0]; r[0] = r[i]; r[i] = t;
+        qrs_heapify(r, i);
+    }
+}
+
+/* ------------------------------------------------------------------ *
+ * visited set — per-query global open-addressing table, atomicCAS
+ * claim (set membership only; probe order is not part of the parity
+ * contract — OptHashPosVector is a pure set, WorkSpace.h:113).
+ * ------------------------------------------------------------------ */
+
+/* returns 1 if idx was already present; inserts otherwise */
+DEV int visited_test_insert(int32_t* tab, uint32_t mask, int32_t idx, int* oflow)
+{
+    uint32_t key = (uint32_t)(idx + 1);
+    uint32_t h = (key * 2654435761u) & mask;
+    for (int probe = 0; probe < 4096; probe++) {
+        int32_t cur = atomicCAS(&tab[h], 0, (int32_t)key);
+        if (cur == 0) return 0;
+        if (cur == (int32_t)key) return 1;
+        h = (h + 1) & mask;
+    }
+    *oflow = 1;
+    return 1;
+}
+
+/* ------------------------------------------------------------------ *
+ * distances — one candidate per 16-lane group, reference order
+ * ------------------------------------------------------------------ */
+
+/* float: the AVX512 chunk/fold order of DistanceUtils.cpp:650 (L2) /
+ * the cosine analog, with fused lane accumulate (fmaf) as compiled in
+ * the reference build (see oracle/sptag_oracle.c header). Result valid
+ * on every lane of the group. */
+template <int DM>
+DEV float ref_dist_grp_f32(const float* __restrict__ q,
+                           const float* __restrict__ v, int d)
+{
+    const int g = threadIdx.x & 15;
+    float a = 0.0f;
+    const int nd16 = (d >> 4) << 4;
+    for (int i = 0; i < nd16; i += 16) {
+        float x = q[i + g], y = v[i + g];
+        a = (DM == DM_L2) ? fmaf(x - y, x - y, a) : fmaf(x, y, a);
+    }
+    a = a + __shfl_down(a, 8, 16);           /* a8[g] on lanes g<8 */
+    const int nd8 = (d >> 3) << 3;
+    for (int i = nd16; i < nd8; i += 8) {
+        if (g < 8) {
+            float x = q[i + g], y = v[i + g];
+            a = (DM == DM_L2) ? fmaf(x - y, x - y, a) : fmaf(x, y, a);
+        }
+    }
+    a = a + __shfl_down(a, 4, 16);           /* a4[g] on lanes g<4 */
+    const int nd4 = (d >> 2) << 2;
+    for (int i = nd8; i < nd4; i += 4) {
+        if (g < 4) {
+            float x = q[i + g], y = v[i + g];
+            a = (DM == DM_L2) ? fmaf(x - y, x - y, a) : fmaf(x, y, a);
+        }
+    }
+    float s = ((__shfl(a, 0, 16) + __shfl(a, 1, 16)) + __shfl(a, 2, 16)) + __shfl(a, 3, 16);
+    for (int i = nd4; i < d; i++) {
+        float x = q[i], y = v[i];
+        s = (DM == DM_L2) ? fmaf(x - y, x - y, s) : fmaf(x, y, s);
+    }
+    return (DM == DM_L2) ? s : 1.0f - s;
+}
+
+/* int8: exact integer accumulation (order-free; partials exact in the
+ * reference's float lanes for dim*254^2 < 2^24 — guarded host-side). */
+template <int DM>
+DEV float ref_dist_grp_i8(const int8_t* __restrict__ q,
+                          const int8_t* __restrict__ v, int d)
+{
+    const int g = threadIdx.x & 15;
+    int s = 0;
+    if ((d & 3) == 0) {
+        const int W = d >> 2;
+        const int32_t* qw = (const int32_t*)q;
+        const int32_t* vw = (const int32_t*)v;
+        for (int w = g; w < W; w += 16) {
+            int32_t a = qw[w], b = vw[w];
+#pragma unroll
+            for (int byte = 0; byte < 4; byte++) {
+                int xa = (int)(int8_t)(a >> (8 * byte));
+                int xb = (int)(int8_t)(b >> (8 * byte));
+                s += (DM == DM_L2) ? (xa - xb) * (xa - xb) : xa * xb;
+            }
+        }
+    } else {
+        for (int i = g; i < d; i += 16) {
+            int xa = q[i], xb = v[i];
+            s += (DM == DM_L2) ? (xa - xb) * (xa - xb) : xa * xb;
+        }
+    }
+    s += __shfl_xor(s, 8, 16);
+    s += __shfl_xor(s, 4, 16);
+    s += __shfl_xor(s, 2, 16);
+    s += __shfl_xor(s, 1, 16);
+    return (DM == DM_L2) ? (float)s : (float)(16129 - s);
+}
+
+template <typename T, int DM>
+DEV float ref_dist_grp(const T* q, const T* v, int d);
+template <> DEV float ref_dist_grp<float, DM_L2>(const float* q, const float* v, int d)
+{ return ref_dist_grp_f32<DM_L2>(q, v, d); }
+template <> DEV float ref_dist_grp<float, DM_COSINE>(const float* q, const float* v, int d)
+{ return ref_dist_grp_f32<DM_COSINE>(q, v, d); }
+template <> DEV float ref_dist_grp<int8_t, DM_L2>(const int8_t* q, const int8_t* v, int d)
+{ return ref_dist_grp_i8<DM_L2>(q, v, d); }
+template <> DEV float ref_dist_grp<int8_t, DM_COSINE>(const int8_t* q, const int8_t* v, int d)
+{ return ref_dist_grp_i8<DM_COSINE>(q, v, d); }
+
+/* ------------------------------------------------------------------ *
+ * per-query context
+ * ------------------------------------------------------------------ */
+
+struct SerialState {          /* in LDS; lane 0 writes, wave reads after sync */
+    int ng_count, spt_count, dpq_len, checked;
+    int oflow, terminate, break_flag, want_tree;
+    NodeDist popped;
+};
+
+template <typename T>
+struct QCtx {
+    const DevIndex* di;
+    const SearchCfg* cfg;
+    const T* qlds;            /* query vector staged in LDS */
+    float* dstage;            /* MAX_DEG staged distances */
+    int32_t* istage;          /* MAX_DEG staged ids / compacted lanes */
+    QRes* qrs;                /* k results */
+    float* dpq;               /* dpq_cap+1 */
+    HeapRef ng, spt;
+    int32_t* vtab;            /* this query's visited table */
+    uint32_t vmask;
+    SerialState* ss;
+    int lane;
+};
+
+template <typename T>
+DEV const T* vec_at(const DevIndex& di, int32_t v)
+{
+    return (const T*)di.vectors + (size_t)v * di.dim;
+}
+
+DEV int not_deleted(const DevIndex& di, int32_t v)
+{
+    /* StaticDispatch::CheckIfNotDeleted vs AlwaysTrue (BKTIndex.cpp:437,
+     * dispatch :471-507): deletes ignored when the index has none. */
+    if (!di.has_deleted) return 1;
+    return di.deleted[v] == 0;
+}
+
+/* stage distances of `cnt` (<=64) data vectors whose ids are in istage[0..cnt)
+ * into dstage[0..cnt); whole-wave cooperative, 4 at a time. */
+template <typename T, int DM>
+DEV void stage_dists(QCtx<T>& c, int cnt)
+{
+    for (int r = 0; r < cnt; r += 4) {
+        int j = r + (c.lane >> 4);
+        if (j < cnt) {
+            float dv = ref_dist_grp<T, DM>(c.qlds, vec_at<T>(*c.di, c.istage[j]), c.di->dim);
+            if ((c.lane & 15) == 0) c.dstage[j] = dv;
+        }
+    }
+    __syncthreads();
+}
+
+/* BKTree.h:772 SearchTrees — wave-cooperative, serial decisions on lane 0. */
+template <typename T, int DM>
+DEV void search_trees_dev(QCtx<T>& c, int limit)
+{
+    const DevIndex& di = *c.di;
+    for (;;) {
+        __syncthreads();
+        if (c.ss->spt_count <= 0 || c.ss->oflow) break;
+        if (c.lane == 0) c.ss->popped = ndheap_pop(c.spt, &c.ss->spt_count);
+        __syncthreads();
+        NodeDist bcell = c.ss->popped;
+        int32_t center = di.tree_nodes[(size_t)bcell.node * 3 + 0];
+        int32_t cs = di.tree_nodes[(size_t)bcell.node * 3 + 1];
+        int32_t ce = di.tree_nodes[(size_t)bcell.node * 3 + 2];
+        if (cs < 0) {
+            if (c.lane == 0) {
+                if (!visited_test_insert(c.vtab, c.vmask, center, &c.ss->oflow)) {
+                    c.ss->checked++;
+                    ndheap_insert(c.ng, &c.ss->ng_count, NodeDist{center, bcell.distance}, &c.ss->oflow);
+                }
+                c.ss->break_flag = (c.ss->checked >= limit);
+            }
+            __syncthreads();
+            if (c.ss->break_flag) break;
+        } else {
+            if (c.lane == 0) {
+                if (!visited_test_insert(c.vtab, c.vmask, center, &c.ss->oflow)) {
+                    ndheap_insert(c.ng, &c.ss->ng_count, NodeDist{center, bcell.distance}, &c.ss->oflow);
+                }
+            }
+            for (int base = cs; base < ce; base += 64) {
+                int cnt = min(64, ce - base);
+                if (c.lane < cnt) c.istage[c.lane] = di.tree_nodes[(size_t)(base + c.lane) * 3];
+                __syncthreads();
+                stage_dists<T, DM>(c, cnt);
+                if (c.lane == 0)
+                    for (int i = 0; i < cnt; i++)
+                        ndheap_insert(c.spt, &c.ss->spt_count,
+                                      NodeDist{base + i, c.dstage[i]}, &c.ss->oflow);
+                __syncthreads();
+            }
+        }
+    }
+    __syncthreads();
+}
+
+/* BKTree.h:697 InitSearchTrees (m_bfs = 0). */
+template <typename T, int DM>
+DEV void init_search_trees_dev(QCtx<T>& c)
+{
+    const DevIndex& di = *c.di;
+    for (int t = 0; t < di.ntrees; t++) {
+        int32_t root = di.tree_start[t];
+        int32_t center = di.tree_nodes[(size_t)root * 3 + 0];
+        int32_t cs = di.tree_nodes[(size_t)root * 3 + 1];
+        int32_t ce = di.tree_nodes[(size_t)root * 3 + 2];
+        if (cs < 0) {
+            if (c.lane < 16) {
+                float dv = ref_dist_grp<T, DM>(c.qlds, vec_at<T>(di, center), di.dim);
+                if (c.lane == 0)
+                    ndheap_insert(c.spt, &c.ss->spt_count, NodeDist{root, dv}, &c.ss->oflow);
+            }
+            __syncthreads();
+        } else {
+            for (int base = cs; base < ce; base += 64) {
+                int cnt = min(64, ce - base);
+                if (c.lane < cnt) c.istage[c.lane] = di.tree_nodes[(size_t)(base + c.lane) * 3];
+                __syncthreads();
+                stage_dists<T, DM>(c, cnt);
+                if (c.lane == 0)
+                    for (int i = 0; i < cnt; i++)
+                        ndheap_insert(c.spt, &c.ss->spt_count,
+                                      NodeDist{base + i, c.dstage[i]}, &c.ss->oflow);
+                __syncthreads();
+            }
+        }
+    }
+}
+
+/* ------------------------------------------------------------------ *
+ * main search kernel
+ * ------------------------------------------------------------------ */
+
+template <typename T, int DM, bool LDSHEAP>
+__global__ __launch_bounds__(64)
+void bkt_search_kernel(DevIndex di, SearchCfg cfg, SearchBufs bufs)
+{
+    const int q = blockIdx.x;
+    if (q >= cfg.nq) return;
+    const int lane = threadIdx.x;
+
+    extern __shared__ char smem[];
+    size_t off = 0;
+    T* qlds = (T*)(smem + off);
+    off += ((size_t)di.dim * sizeof(T) + 15) & ~15ul;
+    float* dstage = (float*)(smem + off); off += MAX_DEG * 4;
+    int32_t* istage = (int32_t*)(smem + off); off += MAX_DEG * 4;
+    QRes* qrs = (QRes*)(smem + off); off += (size_t)cfg.k * 8;
+    float* dpq = (float*)(smem + off); off += ((size_t)cfg.dpq_cap + 1) * 4;
+    SerialState* ss = (SerialState*)(smem + off); off += 64;
+
+    HeapRef ng, spt;
+    ng.cap = cfg.ng_cap;  ng.ref_cap = cfg.max_check * 30;   /* WorkSpace.h:265 */
+    spt.cap = cfg.spt_cap; spt.ref_cap = cfg.max_check * 10;
+    if (LDSHEAP) {
+        ng.a = (NodeDist*)(smem + off); off += ((size_t)cfg.ng_cap + 1) * 8;
+        spt.a = (NodeDist*)(smem + off); off += ((size_t)cfg.spt_cap + 1) * 8;
+    } else {
+        ng.a = (NodeDist*)bufs.gheap_ng + (size_t)q * (cfg.ng_cap + 1);
+        spt.a = (NodeDist*)bufs.gheap_spt + (size_t)q * (cfg.spt_cap + 1);
+    }
+
+    /* stage query */
+    const T* gq = (const T*)bufs.queries + (size_t)q * di.dim;
+    for (int i = lane; i < di.dim; i += 64) qlds[i] = gq[i];
+
+    if (lane == 0) {
+        ss->ng_count = 0; ss->spt_count = 0; ss->dpq_len = 1; ss->checked = 0;
+        ss->oflow = 0; ss->terminate = 0; ss->break_flag = 0; ss->want_tree = 0;
+        ng.a[0] = NodeDist{-1, MAXDIST};   /* Heap empty-top sentinel */
+        spt.a[0] = NodeDist{-1, MAXDIST};
+        dpq[1] = MAXDIST;                  /* DistPriorityQueue seed */
+        for (int i = 0; i < cfg.k; i++) qrs[i] = QRes{-1, MAXDIST};
+    }
+    __syncthreads();
+
+    QCtx<T> c{&di, &cfg, qlds, dstage, istage, qrs, dpq, ng, spt,
+              bufs.visited + (size_t)q * cfg.vcap, (uint32_t)(cfg.vcap - 1),
+              ss, lane};
+
+    init_search_trees_dev<T, DM>(c);
+    __syncthreads();
+    search_trees_dev<T, DM>(c, cfg.init_pivots);
+
+    const int deg = di.deg;
+    const int checkPos = deg - 1;
+
+    for (;;) {
+        __syncthreads();
+        if (ss->ng_count <= 0 || ss->terminate || ss->oflow) break;
+        if (lane == 0) ss->popped = ndheap_pop(c.ng, &ss->ng_count);
+        __syncthreads();
+        NodeDist gnode = ss->popped;
+        const int32_t* row = di.graph + (size_t)gnode.node * deg;
+        int32_t nn = lane < deg ? row[lane] : -1;
+        uint64_t negm = __ballot(lane >= deg || nn < 0);
+        int firstneg = negm ? (int)__builtin_ctzll(negm) : 64;
+        int32_t checkNode = __shfl(nn, checkPos);
+
+        if (lane == 0) {
+            /* BKTIndex.cpp:290-331: result/termination block */
+            if (gnode.distance <= qrs[0].dist) {
+                if (checkNode < -1) {
+                    /* duplicate-center chain (BKTIndex.cpp:292-312) */
+                    const int32_t* tn = &di.tree_nodes[(size_t)(-2 - checkNode) * 3];
+                    int32_t i = -tn[1];
+                    int32_t tmpNode = gnode.node;
+                    do {
+                        if (not_deleted(di, tmpNode)) {
+                            if (!qrs_add(qrs, cfg.k, tmpNode, gnode.distance)) break;
+                        }
+                        if (i <= 0) break;
+                        tmpNode = di.tree_nodes[(size_t)i * 3];
+                    } while (i++ < tn[2]);
+                } else {
+                    if (not_deleted(di, gnode.node))
+                        qrs_add(qrs, cfg.k, gnode.node, gnode.distance);
+                }
+            } else {
+                if (not_deleted(di, gnode.node)) {
+                    if (gnode.distance > dpq[1] || ss->checked > cfg.max_check)
+                        ss->terminate = 1;
+                }
+            }
+        }
+        __syncthreads();
+        if (ss->terminate) break;
+
+        /* neighbor expansion (BKTIndex.cpp:333-345) */
+        int already = 1;
+        if (lane < firstneg)
+            already = visited_test_insert(c.vtab, c.vmask, nn, &ss->oflow);
+        uint64_t candm = __ballot(lane < firstneg && !already);
+        istage[lane] = nn;
+        int ncand = __popcll(candm);
+        if ((candm >> lane) & 1) {
+            int pos = __popcll(candm & ((1ull << lane) - 1));
+            dstage[pos] = __int_as_float(lane);   /* compacted lane list */
+        }
+        __syncthreads();
+        /* move compacted lane ids out of dstage before distances overwrite */
+        int clane = lane < ncand ? __float_as_int(dstage[lane]) : -1;
+        int32_t cvid = clane >= 0 ? istage[clane] : -1;
+        __syncthreads();
+        for (int r = 0; r < ncand; r += 4) {
+            int j = r + (lane >> 4);
+            int32_t vid = __shfl(cvid, j & 63);   /* candidate j's vid */
+            int32_t cl = __shfl(clane, j & 63);
+            if (j < ncand) {
+                float dv = ref_dist_grp<T, DM>(qlds, vec_at<T>(di, vid), di.dim);
+                if ((lane & 15) == 0) dstage[cl] = dv;
+            }
+        }
+        __syncthreads();
+        if (lane == 0) {
+            for (int i = 0; i < firstneg; i++) {
+                if (!((candm >> i) & 1)) continue;
+                float dv = dstage[i];
+                ss->checked++;
+                if (dpq_insert(dpq, &ss->dpq_len, cfg.dpq_cap, dv))
+                    ndheap_insert(c.ng, &ss->ng_count, NodeDist{istage[i], dv}, &ss->oflow);
+            }
+            /* dynamic pivots (BKTIndex.cpp:346-349) */
+            ss->want_tree = (ndheap_top(c.ng, ss->ng_count).distance >
+                             ndheap_top(c.spt, ss->spt_count).distance);
+        }
+        __syncthreads();
+        if (ss->want_tree)
+            search_trees_dev<T, DM>(c, cfg.other_pivots + ss->checked);
+    }
+    __syncthreads();
+
+    if (lane == 0) {
+        bufs.oflow[q] = ss->oflow;
+        qrs_sort(qrs, cfg.k);
+    }
+    __syncthreads();
+    if (lane < cfg.k) {
+        bufs.out_vids[(size_t)q * cfg.k + lane] = qrs[lane].vid;
+        bufs.out_dists[(size_t)q * cfg.k + lane] = qrs[lane].dist;
+    }
+}
+
+/* ------------------------------------------------------------------ *
+ * brute-force truth kernel (TruthSet::GenerateTruth semantics,
+ * TruthSet.h:163): exact top-k by (dist, vid) over all non-deleted rows.
+ * One query per wave; test-scale only.
+ * ------------------------------------------------------------------ */
+
+template <typename T, int DM>
+__global__ __launch_bounds__(64)
+void truth_kernel(DevIndex di, const void* queries, int32_t nq, int32_t k,
+                  int32_t* out_vids, float* out_dists)
+{
+    const int q = blockIdx.x;
+    if (q >= nq) return;
+    const int lane = threadIdx.x;
+
+    extern __shared__ char smem[];
+    size_t off = 0;
+    T* qlds = (T*)(smem + off);
+    off += ((size_t)di.dim * sizeof(T) + 15) & ~15ul;
+    QRes* qrs = (QRes*)(smem + off); off += (size_t)k * 8;
+    float* dstage = (float*)(smem + off);
+
+    const T* gq = (const T*)queries + (size_t)q * di.dim;
+    for (int i = lane; i < di.dim; i += 64) qlds[i] = gq[i];
+    if (lane == 0)
+        for (int i = 0; i < k; i++) qrs[i] = QRes{-1, MAXDIST};
+    __syncthreads();
+
+    for (int32_t base = 0; base < di.n; base += 4) {
+        int cnt = min(4, di.n - base);
+        int j = lane >> 4;
+        if (j < cnt) {
+            float dv = ref_dist_grp<T, DM>(qlds, vec_at<T>(di, base + j), di.dim);
+            if ((lane & 15) == 0) dstage[j] = dv;
+        }
+        __syncthreads();
+        if (lane == 0) {
+            for (int i = 0; i < cnt; i++) {
+                if (di.has_deleted && di.deleted[base + i]) continue;
+                qrs_add(qrs, k, base + i, dstage[i]);
+            }
+        }
+        __syncthreads();
+    }
+    if (lane == 0) qrs_sort(qrs, k);
+    __syncthreads();
+    if (lane < k) {
+        out_vids[(size_t)q * k + lane] = qrs[lane].vid;
+        out_dists[(size_t)q * k + lane] = qrs[lane].dist;
+    }
+}
+
+/* ------------------------------------------------------------------ *
+ * launchers
+ * ------------------------------------------------------------------ */
+
+template <typename T, int DM, bool LDSHEAP>
+static int launch_one(const DevIndex& di, const SearchCfg& cfg,
+                      const SearchBufs& bufs, hipStream_t stream)
+{
+    size_t lds = lds_bytes(di.dim, sizeof(T), cfg, LDSHEAP);
+    dim3 grid(cfg.nq), block(64);
+    hipLaunchKernelGGL((bkt_search_kernel<T, DM, LDSHEAP>), grid, block, lds,
+                       stream, di, cfg, bufs);
+    return (int)hipGetLastError();
+}
+
+int launch_bkt_search(int vt, int dm, bool heaps_in_lds, const DevIndex& di,
+                      const SearchCfg& cfg, const SearchBufs& bufs, void* stream)
+{
+    hipStream_t s = (hipStream_t)stream;
+    if (vt == VT_FLOAT && dm == DM_L2)
+        return heaps_in_lds ? launch_one<float, DM_L2, true>(di, cfg, bufs, s)
+                            : launch_one<float, DM_L2, false>(di, cfg, bufs, s);
+    if (vt == VT_FLOAT && dm == DM_COSINE)
+        return heaps_in_lds ? launch_one<float, DM_COSINE, true>(di, cfg, bufs, s)
+                            : launch_one<float, DM_COSINE, false>(di, cfg, bufs, s);
+    if (vt == VT_INT8 && dm == DM_L2)
+        return heaps_in_lds ? launch_one<int8_t, DM_L2, true>(di, cfg, bufs, s)
+                            : launch_one<int8_t, DM_L2, false>(di, cfg, bufs, s);
+    if (vt == VT_INT8 && dm == DM_COSINE)
+        return heaps_in_lds ? launch_one<int8_t, DM_COSINE, true>(di, cfg, bufs, s)
+                            : launch_one<int8_t, DM_COSINE, false>(di, cfg, bufs, s);
+    return (int)hipErrorInvalidValue;
+}
+
+template <typename T, int DM>
+static int launch_truth_one(const DevIndex& di, const void* queries, int32_t nq,
+                            int32_t k, int32_t* ov, float* od, hipStream_t s)
+{
+    size_t lds = (((size_t)di.dim * sizeof(T) + 15) & ~15ul) + (size_t)k * 8 + 64;
+    hipLaunchKernelGGL((truth_kernel<T, DM>), dim3(nq), dim3(64), lds, s,
+                       di, queries, nq, k, ov, od);
+    return (int)hipGetLastError();
+}
+
+int launch_truth(int vt, int dm, const DevIndex& di, const void* queries,
+                 int32_t nq, int32_t k, int32_t* ov, float* od, void* stream)
+{
+    hipStream_t s = (hipStream_t)stream;
+    if (vt == VT_FLOAT && dm == DM_L2) return launch_truth_one<float, DM_L2>(di, queries, nq, k, ov, od, s);
+    if (vt == VT_FLOAT && dm == DM_COSINE) return launch_truth_one<float, DM_COSINE>(di, queries, nq, k, ov, od, s);
+    if (vt == VT_INT8 && dm == DM_L2) return launch_truth_one<int8_t, DM_L2>(di, queries, nq, k, ov, od, s);
+    if (vt == VT_INT8 && dm == DM_COSINE) return launch_truth_one<int8_t, DM_COSINE>(di, queries, nq, k, ov, od, s);
+    return (int)hipErrorInvalidValue;
+}
+
+}  /* namespace sptag_amd */

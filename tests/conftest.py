@@ -1,0 +1,48 @@
+import json
+import os
+import sys
+
+import numpy as np
+import pytest
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+if REPO not in sys.path:
+    sys.path.insert(0, REPO)
+
+GOLDEN = os.path.join(REPO, "tests", "golden")
+
+
+def pytest_configure(config):
+    config.addinivalue_line(
+        "markers", "gpu: needs a HIP device (run on the MI355X box)")
+
+
+def golden_fixtures():
+    names = []
+    if os.path.isdir(GOLDEN):
+        for name in sorted(os.listdir(GOLDEN)):
+            if os.path.isdir(os.path.join(GOLDEN, name)):
+                names.append(name)
+    return names
+
+
+def load_golden(name):
+    d = os.path.join(GOLDEN, name)
+    meta = json.load(open(os.path.join(d, "meta.json")))
+    dtype = np.float32 if meta["valuetype"] == "Float" else np.int8
+    with open(os.path.join(d, "queries.bin"), "rb") as f:
+        n, dim = np.frombuffer(f.read(8), dtype=np.int32)
+        queries = np.frombuffer(f.read(), dtype=dtype).reshape(n, dim)
+    results = {}
+    for mc in meta["maxchecks"]:
+        raw = open(os.path.join(d, f"results_mc{mc}.bin"), "rb").read()
+        rec = np.frombuffer(raw[8:], dtype=np.dtype([("vid", np.int32),
+                                                     ("dist", np.float32)]))
+        results[mc] = rec.reshape(meta["nq"], meta["k"])
+    return {"dir": d, "index": os.path.join(d, "index"), "meta": meta,
+            "queries": queries, "results": results}
+
+
+@pytest.fixture(scope="session")
+def repo_root():
+    return REPO
