@@ -87,6 +87,21 @@ int sptag_amd_search_batch(SptagAmdIndex* idx, const void* queries, int32_t nq,
                            int32_t k, int32_t max_check,
                            int32_t* out_vids, float* out_dists);
 
+/* Device-pointer variant: queries/outputs are HIP device buffers already
+ * resident in HBM (e.g. torch tensors' data_ptr); no host<->device copies
+ * inside the timed path. Same contract otherwise. */
+int sptag_amd_search_batch_device(SptagAmdIndex* idx, const void* d_queries,
+                                  int32_t nq, int32_t k, int32_t max_check,
+                                  int32_t* d_out_vids, float* d_out_dists);
+
+/* Instrumentation of the LAST search call on this handle: accumulated HIP
+ * kernel time (events around each search launch, on the launch stream) and
+ * the traversal totals across all queries (checked = distance evaluations,
+ * popped = frontier pops) — the inputs to the roofline's algorithmic-bytes
+ * numerator. */
+void sptag_amd_last_stats(SptagAmdIndex* idx, double* kernel_ms,
+                          long long* checked, long long* popped);
+
 /* Exact brute-force top-k on the GPU (truth generation / recall gates;
  * reference TruthSet::GenerateTruth semantics, inc/Core/Common/TruthSet.h:163). */
 int sptag_amd_truth(SptagAmdIndex* idx, const void* queries, int32_t nq,

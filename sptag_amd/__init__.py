@@ -59,6 +59,11 @@ def load_library():
     lib.sptag_amd_search_batch.argtypes = [
         ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int32, ctypes.c_int32,
         ctypes.c_int32, ctypes.c_void_p, ctypes.c_void_p]
+    lib.sptag_amd_search_batch_device.restype = ctypes.c_int
+    lib.sptag_amd_search_batch_device.argtypes = lib.sptag_amd_search_batch.argtypes
+    lib.sptag_amd_last_stats.argtypes = [
+        ctypes.c_void_p, ctypes.POINTER(ctypes.c_double),
+        ctypes.POINTER(ctypes.c_longlong), ctypes.POINTER(ctypes.c_longlong)]
     lib.sptag_amd_truth.restype = ctypes.c_int
     lib.sptag_amd_truth.argtypes = lib.sptag_amd_search_batch.argtypes[:4] + [
         ctypes.c_void_p, ctypes.c_void_p]
@@ -182,6 +187,25 @@ class AnnIndex:
         if rc != 0:
             raise SptagAmdError(rc, "search_batch")
         return vids, dists
+
+    def BatchSearchDevice(self, d_queries_ptr, nq, k, d_vids_ptr, d_dists_ptr,
+                          max_check=0):
+        """Device-resident search: all pointers are HIP device addresses
+        (e.g. torch cuda tensors' .data_ptr()); no PCIe copies inside."""
+        rc = self._lib.sptag_amd_search_batch_device(
+            self._h, ctypes.c_void_p(d_queries_ptr), nq, k, max_check,
+            ctypes.c_void_p(d_vids_ptr), ctypes.c_void_p(d_dists_ptr))
+        if rc != 0:
+            raise SptagAmdError(rc, "search_batch_device")
+
+    def LastStats(self):
+        """(kernel_ms, checked, popped) of the last search call."""
+        ms = ctypes.c_double()
+        ch = ctypes.c_longlong()
+        po = ctypes.c_longlong()
+        self._lib.sptag_amd_last_stats(self._h, ctypes.byref(ms),
+                                       ctypes.byref(ch), ctypes.byref(po))
+        return ms.value, ch.value, po.value
 
     def Search(self, query, k, max_check=0):
         vids, dists = self.BatchSearch(query, k, max_check)

@@ -1,0 +1,419 @@
+"""GPU-accelerated index builder (torch tooling over the GPU for the heavy
+linear algebra; rocBLAS GEMMs via torch.matmul).
+
+Mirrors the reference build SEMANTICS (citations to /root/reference/AnnService):
+  - cosine pre-normalization of base vectors to norm=base in place
+    (BKTIndex.cpp:749-756, CommonUtils.h:62 — C-cast truncation for int8);
+  - BKT: hierarchical k-means tree, node children contiguous, leaf nodes are
+    the vectors themselves, node centerid is a member vector
+    (BKTree.h:547-625). Build is not deterministic in the reference either
+    (std::mt19937 shuffles — SURVEY.md §8c), so equivalence is judged by
+    search recall/QPS on the produced index, not bytes.
+  - KNN graph: TP-tree partitions, exact KNN inside each leaf, candidates
+    merged over trees (NeighborhoodGraph.h:301-360);
+  - RNG prune: candidates ascending by distance, accept c iff for every
+    already-accepted b: RNGFactor*dist(b,c) >= dist(q,c), up to degree, pad
+    -1 (RelativeNeighborhoodGraph.h:18-35 RebuildNeighbors).
+
+Output arrays load via AnnIndex.FromArrays / sptag_amd_create_index and
+save to the reference's byte format via AnnIndex.Save.
+"""
+import numpy as np
+import torch
+
+
+def _dev():
+    return "cuda" if torch.cuda.is_available() else "cpu"
+
+
+def normalize_base(vectors, distmethod):
+    """Reference cosine build normalization (BKTIndex.cpp:749-756)."""
+    if distmethod != "Cosine":
+        return vectors
+    x = vectors.astype(np.float64)
+    norms = np.sqrt((x * x).sum(axis=1, keepdims=True))
+    norms[norms == 0] = 1.0
+    if vectors.dtype == np.int8:
+        out = np.trunc(x * 127.0 / norms)  # C-cast truncation (CommonUtils.h:62)
+        return out.astype(np.int8)
+    return (x / norms).astype(np.float32)
+
+
+# ------------------------------------------------------------------ #
+# BKT tree
+# ------------------------------------------------------------------ #
+
+def _kmeans_split(xf, members, K, iters=8, sample=4096, gen=None):
+    """k-means split of one cluster. Returns (groups: list of int64 tensors,
+    reps: list of representative vector ids)."""
+    m = members.numel()
+    K = min(K, max(2, m // 8))
+    if m > sample:
+        sel = torch.randperm(m, generator=gen, device=members.device)[:sample]
+        smp = members[sel]
+    else:
+        smp = members
+    pts = xf[smp]
+    perm = torch.randperm(smp.numel(), generator=gen, device=members.device)[:K]
+    centers = pts[perm].clone()
+    for _ in range(iters):
+        lab = torch.cdist(pts, centers).argmin(1)
+        sums = torch.zeros_like(centers)
+        sums.index_add_(0, lab, pts)
+        cnt = torch.bincount(lab, minlength=centers.shape[0]).clamp(min=1)
+        newc = sums / cnt[:, None].float()
+        keep = torch.bincount(lab, minlength=centers.shape[0]) > 0
+        centers = torch.where(keep[:, None], newc, centers)
+    xm = xf[members]
+    d2 = torch.cdist(xm, centers)          # [m, K]
+    lab = d2.argmin(1)
+    Kc = centers.shape[0]
+    # representative per cluster = member closest to the centroid (vectorized)
+    big = torch.finfo(d2.dtype).max
+    dmask = d2.clone()
+    onehot = torch.nn.functional.one_hot(lab, Kc).bool()
+    dmask[~onehot] = big
+    rep_local = dmask.argmin(0)            # [K]
+    counts = torch.bincount(lab, minlength=Kc)
+    groups, reps = [], []
+    order = lab.argsort(stable=True)
+    start = 0
+    for k in range(Kc):
+        c = int(counts[k].item())
+        if c == 0:
+            continue
+        groups.append(members[order[start:start + c]])
+        reps.append(int(members[rep_local[k]].item()))
+        start += c
+    if len(groups) <= 1:
+        groups, reps = [], []
+        per = (m + K - 1) // K
+        for k in range(0, m, per):
+            g = members[k:k + per]
+            groups.append(g)
+            reps.append(int(g[0].item()))
+    return groups, reps
+
+
+def build_bkt_tree(vectors, *, kmeans_k=32, leaf_size=32, big_cluster=2048,
+                   seed=2016, device=None, verbose=False):
+    """Returns (tree_start int32[ntrees], tree_nodes int32[N,3]) in the
+    reference layout (BKTree.h:25 BKTNode {centerid,childStart,childEnd},
+    children contiguous, childEnd exclusive, leaves childStart=-1)."""
+    device = device or _dev()
+    n = vectors.shape[0]
+    xf = torch.as_tensor(vectors, device=device)
+    if xf.dtype != torch.float32:
+        xf = xf.float()
+    gen = torch.Generator(device=device)
+    gen.manual_seed(seed)
+
+    cent, cs, ce = [n], [-1], [-1]   # root: centerid=n (unused; as reference)
+    small = []                        # (node_idx, members) clusters <= big_cluster
+    stack = [(0, torch.arange(n, device=device, dtype=torch.int64))]
+    while stack:
+        node, members = stack.pop()
+        groups, reps = _kmeans_split(xf, members, kmeans_k, gen=gen)
+        cs[node] = len(cent)
+        for g, r in zip(groups, reps):
+            child = len(cent)
+            cent.append(r)
+            cs.append(-1)
+            ce.append(-1)
+            if g.numel() > big_cluster:
+                stack.append((child, g))
+            elif g.numel() > 1:
+                small.append((child, g))
+        ce[node] = len(cent)
+        if verbose and len(cent) % 100000 < kmeans_k:
+            print(f"  bkt phase A: {len(cent)} nodes, stack {len(stack)}")
+
+    cent = np.array(cent, dtype=np.int64)
+    cs = np.array(cs, dtype=np.int64)
+    ce = np.array(ce, dtype=np.int64)
+
+    # vectorized bottom: each small cluster -> groups of <= leaf_size members
+    # (ordered by distance to the cluster centroid), each group an internal
+    # node whose children are member leaves.
+    if small:
+        nbase = len(cent)
+        sizes = np.array([g.numel() for _, g in small])
+        gcnts = np.ceil(sizes / leaf_size).astype(np.int64)
+        extra = np.where(sizes > leaf_size, gcnts, 0) + sizes
+        offs = nbase + np.concatenate([[0], np.cumsum(extra)[:-1]])
+        total = int(extra.sum())
+        cent2 = np.empty(total, dtype=np.int64)
+        cs2 = np.full(total, -1, dtype=np.int64)
+        ce2 = np.full(total, -1, dtype=np.int64)
+        for (node, g), size, gcnt, off in zip(small, sizes, gcnts, offs):
+            xg = xf[g]
+            order = torch.cdist(xg, xg.mean(0, keepdim=True)).squeeze(1).argsort()
+            gi = g[order].cpu().numpy()
+            p = off - nbase
+            if size <= leaf_size:
+                cs[node] = off
+                ce[node] = off + size
+                cent2[p:p + size] = gi
+            else:
+                cs[node] = off
+                ce[node] = off + gcnt
+                lp = p + gcnt              # local cursor for leaf blocks
+                for gidx in range(int(gcnt)):
+                    mem = gi[gidx * leaf_size:(gidx + 1) * leaf_size]
+                    cent2[p + gidx] = mem[0]
+                    cs2[p + gidx] = nbase + lp
+                    ce2[p + gidx] = nbase + lp + len(mem)
+                    cent2[lp:lp + len(mem)] = mem
+                    lp += len(mem)
+        cent = np.concatenate([cent, cent2])
+        cs = np.concatenate([cs, cs2])
+        ce = np.concatenate([ce, ce2])
+
+    # sentinel (BKTree.h:625)
+    cent = np.concatenate([cent, [-1]])
+    cs = np.concatenate([cs, [-1]])
+    ce = np.concatenate([ce, [-1]])
+
+    tree_nodes = np.stack([cent, cs, ce], axis=1).astype(np.int32)
+    tree_start = np.array([0], dtype=np.int32)
+    return tree_start, tree_nodes
+
+
+# ------------------------------------------------------------------ #
+# TP-tree KNN candidates + RNG prune
+# ------------------------------------------------------------------ #
+
+def _tpt_leaves(xf, leaf, gen):
+    """Random-hyperplane tree: returns (perm, [(start, count), ...]) — leaves
+    are contiguous slices of perm."""
+    n, d = xf.shape
+    perm = torch.arange(n, device=xf.device)
+    seg = torch.zeros(n, dtype=torch.int64, device=xf.device)
+    nseg = 1
+    max_size = n
+    while max_size > leaf:
+        r = torch.randn(d, generator=gen, device=xf.device)
+        proj = xf[perm] @ r
+        o1 = proj.argsort()
+        o2 = seg[o1].argsort(stable=True)
+        order = o1[o2]
+        perm = perm[order]
+        seg = seg[order]                      # segments contiguous, proj-sorted
+        counts = torch.bincount(seg, minlength=nseg)
+        starts = torch.cumsum(counts, 0) - counts
+        half = counts // 2
+        pos = torch.arange(n, device=xf.device) - starts.repeat_interleave(counts)
+        within_hi = pos >= half.repeat_interleave(counts)
+        # only split segments still above the leaf size
+        splitting = (counts > leaf).repeat_interleave(counts)
+        seg = seg * 2 + (within_hi & splitting).long()
+        uniq, seg = torch.unique(seg, return_inverse=True)
+        nseg = uniq.numel()
+        max_size = int(torch.bincount(seg, minlength=nseg).max().item())
+    counts = torch.bincount(seg, minlength=nseg)
+    starts = torch.cumsum(counts, 0) - counts
+    return perm, list(zip(starts.tolist(), counts.tolist()))
+
+
+def _leaf_knn(xf, perm, bounds, kper, chunk_rows=1_000_000):
+    """Exact KNN inside each leaf. Returns (ids int32 [n,kper], dists f32) in
+    original vector-id space (L2 on the f32 view; candidate generation only —
+    exact metric distances are recomputed by the searcher)."""
+    n = xf.shape[0]
+    ids = torch.full((n, kper), -1, dtype=torch.int32, device=xf.device)
+    dst = torch.full((n, kper), float("inf"), device=xf.device)
+    maxleaf = max(c for _, c in bounds)
+    batch = max(1, chunk_rows // maxleaf)
+    starts_t = torch.tensor([s for s, _ in bounds], device=xf.device)
+    counts_t = torch.tensor([c for _, c in bounds], device=xf.device)
+    for i in range(0, len(bounds), batch):
+        st = starts_t[i:i + batch]
+        ct = counts_t[i:i + batch]
+        B = st.numel()
+        m = maxleaf
+        pos = torch.arange(m, device=xf.device)[None, :].expand(B, m)
+        mask = pos < ct[:, None]
+        gidx = (st[:, None] + pos).clamp(max=n - 1)
+        gather = perm[gidx]                    # [B, m]
+        gather = torch.where(mask, gather, torch.zeros_like(gather))
+        pts = xf[gather]                       # [B, m, d]
+        sq = (pts * pts).sum(-1)
+        d2 = sq[:, :, None] + sq[:, None, :] - 2.0 * torch.bmm(pts, pts.transpose(1, 2))
+        d2.diagonal(dim1=1, dim2=2).fill_(float("inf"))
+        d2.masked_fill_(~mask[:, None, :], float("inf"))
+        k = min(kper, m - 1)
+        vals, loc = torch.topk(d2, k, dim=2, largest=False)
+        nbr = torch.gather(gather[:, None, :].expand(B, m, m), 2, loc)
+        rows = gather[mask]
+        ids[rows, :k] = nbr[mask].int()
+        dst[rows, :k] = vals[mask]
+    return ids, dst
+
+
+def _merge_candidates(ids_a, dst_a, ids_b, dst_b, cand, self_ids):
+    """Merge two candidate lists per point: dedupe ids, keep `cand` nearest.
+    All id tensors int32; invalid = -1/inf."""
+    cid = torch.cat([ids_a, ids_b], dim=1)
+    cdd = torch.cat([dst_a, dst_b], dim=1)
+    # drop self and duplicates (keep nearest occurrence)
+    order = cdd.argsort(dim=1, stable=True)
+    cid = torch.gather(cid, 1, order)
+    cdd = torch.gather(cdd, 1, order)
+    sid, sorder = cid.sort(dim=1, stable=True)
+    dup_sorted = torch.zeros_like(sid, dtype=torch.bool)
+    dup_sorted[:, 1:] = sid[:, 1:] == sid[:, :-1]
+    dup = torch.zeros_like(dup_sorted)
+    dup.scatter_(1, sorder, dup_sorted)
+    invalid = dup | (cid < 0) | (cid == self_ids[:, None])
+    cdd = cdd.masked_fill(invalid, float("inf"))
+    order = cdd.argsort(dim=1, stable=True)[:, :cand]
+    return torch.gather(cid, 1, order), torch.gather(cdd, 1, order)
+
+
+def build_rng_graph(vectors, *, degree=32, ntrees=8, tpt_leaf=1000, cand=64,
+                    rng_factor=1.0, seed=2016, device=None, point_chunk=100_000,
+                    verbose=False):
+    """Returns graph int32 [n, degree] (RNG-pruned, ascending, -1 padded)."""
+    device = device or _dev()
+    n = vectors.shape[0]
+    xf = torch.as_tensor(vectors, device=device)
+    if xf.dtype != torch.float32:
+        xf = xf.float()
+    gen = torch.Generator(device=device)
+    gen.manual_seed(seed + 77)
+    self_ids = torch.arange(n, device=device, dtype=torch.int32)
+
+    kper = min(cand // 2 + 1, 48)
+    ids = dst = None
+    for t in range(ntrees):
+        perm, bounds = _tpt_leaves(xf, tpt_leaf, gen)
+        tids, tdst = _leaf_knn(xf, perm, bounds, kper)
+        if ids is None:
+            ids, dst = _merge_candidates(tids, tdst, tids[:, :0], tdst[:, :0],
+                                         cand, self_ids)
+        else:
+            ids, dst = _merge_candidates(ids, dst, tids, tdst, cand, self_ids)
+        del tids, tdst
+        if verbose:
+            print(f"  tpt tree {t + 1}/{ntrees} merged ({len(bounds)} leaves)")
+
+    graph = torch.full((n, degree), -1, dtype=torch.int32, device=device)
+    for s in range(0, n, point_chunk):
+        e = min(n, s + point_chunk)
+        graph[s:e] = _rng_prune(xf, ids[s:e], dst[s:e], degree, rng_factor, device)
+        if verbose and (s // point_chunk) % 20 == 0:
+            print(f"  rng prune {e}/{n}")
+    return graph, ids, dst
+
+
+def _rng_prune(xf, cid, cdd, degree, rng_factor, device):
+    """RNG prune one chunk of candidate lists (ascending by dist)."""
+    C = cid.shape[1]
+    cvec = xf[cid.clamp(min=0).long()]
+    csq = (cvec * cvec).sum(-1)
+    P = csq[:, :, None] + csq[:, None, :] - 2.0 * torch.bmm(cvec, cvec.transpose(1, 2))
+    valid = torch.isfinite(cdd)
+    acc = torch.zeros_like(valid)
+    count = torch.zeros(cid.shape[0], dtype=torch.int32, device=device)
+    for j in range(C):
+        viol = (rng_factor * P[:, :, j] < cdd[:, j:j + 1]) & acc
+        good = valid[:, j] & ~viol.any(1) & (count < degree)
+        acc[:, j] = good
+        count += good.int()
+    out = torch.full((cid.shape[0], degree), -1, dtype=torch.int32, device=device)
+    pos = (torch.cumsum(acc.int(), dim=1) - 1).clamp(min=0)
+    rows = torch.nonzero(acc, as_tuple=True)
+    out[rows[0], pos[rows]] = cid[rows]
+    return out
+
+
+def refine_graph(vectors, graph, cand_ids, cand_dst, *, degree=32, cand=64,
+                 rounds=2, hop_sample=8, rng_factor=1.0, device=None,
+                 point_chunk=100_000, seed=2016, verbose=False):
+    """Neighborhood refinement: the reference refines each node's edges from
+    a CEF-sized candidate pool gathered by searching the index itself
+    (NeighborhoodGraph.h:460-560 RefineGraph/RefineNode). Here the pool is
+    grown NN-descent style — current candidates + two-hop neighbors +
+    reverse edges — then RNG-pruned with the same RebuildNeighbors rule.
+    Returns (graph int32 [n, degree], cand_ids, cand_dst)."""
+    device = device or _dev()
+    n = vectors.shape[0]
+    xf = torch.as_tensor(vectors, device=device)
+    if xf.dtype != torch.float32:
+        xf = xf.float()
+    g = torch.as_tensor(graph, device=device)
+    self_ids = torch.arange(n, device=device, dtype=torch.int32)
+    gen = torch.Generator(device=device)
+    gen.manual_seed(seed + 123)
+
+    for r in range(rounds):
+        # reverse edges (sampled): every edge (i -> j) proposes i to j
+        src = self_ids.repeat_interleave(g.shape[1])
+        dstv = g.reshape(-1)
+        keep = dstv >= 0
+        src, dstv = src[keep], dstv[keep]
+        order = torch.argsort(dstv, stable=True)
+        src, dstv = src[order], dstv[order]
+        cnt = torch.bincount(dstv.long(), minlength=n)
+        start = torch.cumsum(cnt, 0) - cnt
+        rcap = 16
+        pos = torch.arange(rcap, device=device)
+        ridx = (start[:, None] + pos[None, :]).clamp(max=max(len(src) - 1, 0))
+        rmask = pos[None, :] < cnt[:, None]
+        rev = torch.full((n, rcap), -1, dtype=torch.int32, device=device)
+        if len(src):
+            rev[rmask] = src[ridx[rmask]]
+        for s in range(0, n, point_chunk):
+            e = min(n, s + point_chunk)
+            B = e - s
+            nbr = g[s:e]                                   # [B, deg]
+            hop2 = g[nbr[:, :hop_sample].clamp(min=0).long()].reshape(B, -1)
+            hop2 = torch.where((nbr[:, :hop_sample] < 0).repeat_interleave(
+                g.shape[1], dim=1), torch.full_like(hop2, -1), hop2)
+            newc = torch.cat([hop2, rev[s:e]], dim=1)
+            newd = _exact_l2(xf, torch.arange(s, e, device=device), newc)
+            cand_ids[s:e], cand_dst[s:e] = _merge_candidates(
+                cand_ids[s:e], cand_dst[s:e], newc, newd, cand, self_ids[s:e])
+            g[s:e] = _rng_prune(xf, cand_ids[s:e], cand_dst[s:e], degree,
+                                rng_factor, device)
+        if verbose:
+            print(f"  refine round {r + 1}/{rounds} done")
+    return g.cpu().numpy(), cand_ids, cand_dst
+
+
+def _exact_l2(xf, q_rows, c_ids):
+    """exact (f32 view) L2 distances between rows q_rows and candidate ids
+    c_ids [B, C]; invalid ids (<0) -> inf."""
+    a = xf[q_rows]                          # [B, d]
+    b = xf[c_ids.clamp(min=0).long()]       # [B, C, d]
+    diff = b - a[:, None, :]
+    d = (diff * diff).sum(-1)
+    return d.masked_fill(c_ids < 0, float("inf"))
+
+
+def build_index_arrays(vectors, distmethod, *, degree=32, ntrees=8,
+                       tpt_leaf=1000, cand=64, kmeans_k=32, leaf_size=32,
+                       refine_rounds=2, seed=2016, device=None,
+                       normalized=False, verbose=False):
+    """Full build: returns dict(vectors, tree_start, tree_nodes, graph) ready
+    for AnnIndex.FromArrays (vectors already cosine-normalized when needed,
+    as the reference stores them on disk)."""
+    if not normalized:
+        vectors = normalize_base(vectors, distmethod)
+    tree_start, tree_nodes = build_bkt_tree(
+        vectors, kmeans_k=kmeans_k, leaf_size=leaf_size, seed=seed,
+        device=device, verbose=verbose)
+    if verbose:
+        print(f"  bkt tree: {len(tree_nodes)} nodes")
+    graph, cids, cdst = build_rng_graph(
+        vectors, degree=degree, ntrees=ntrees, tpt_leaf=tpt_leaf, cand=cand,
+        seed=seed, device=device, verbose=verbose)
+    if refine_rounds > 0:
+        graph, _, _ = refine_graph(vectors, graph, cids, cdst, degree=degree,
+                                   cand=cand, rounds=refine_rounds,
+                                   device=device, seed=seed, verbose=verbose)
+    else:
+        graph = graph.cpu().numpy()
+    return {"vectors": vectors, "tree_start": tree_start,
+            "tree_nodes": tree_nodes, "graph": graph,
+            "distmethod": distmethod}

@@ -41,6 +41,7 @@ struct SearchBufs {
     float* out_dists;      /* nq*k */
     int32_t* visited;      /* nq*vcap, zeroed before launch */
     int32_t* oflow;        /* nq flags: nonzero => rerun with bigger caps */
+    int32_t* stats;        /* nq*2: {checked, popped} per query (may be null) */
     /* only used by the global-heap variant: per-query strided scratch */
     void* gheap_ng;        /* nq * (ng_cap+1) * 8B */
     void* gheap_spt;       /* nq * (spt_cap+1) * 8B */

@@ -50,6 +50,15 @@ struct HostIndex {
     std::vector<uint8_t> h_deleted;
     int64_t deleted_count = 0;
     std::mutex lock;
+    /* cached per-handle search scratch (grown on demand) */
+    int32_t* d_visited = nullptr; size_t visited_cap = 0;
+    int32_t* d_oflow = nullptr;   size_t oflow_cap = 0;
+    int32_t* d_stats = nullptr;   size_t stats_cap = 0;
+    void* d_gng = nullptr;        size_t gng_cap = 0;
+    void* d_gspt = nullptr;       size_t gspt_cap = 0;
+    hipEvent_t ev0 = nullptr, ev1 = nullptr;
+    double last_kernel_ms = 0;
+    long long last_checked = 0, last_popped = 0;
 
     size_t esz() const { return vt == VT_FLOAT ? 4 : 1; }
     DevIndex dev() const {
@@ -286,28 +295,37 @@ SptagAmdIndex* sptag_amd_load_index(const char* folder, int device)
 void sptag_amd_free_index(SptagAmdIndex* ix)
 {
     if (!ix) return;
-    if (ix->d_vectors) hipFree(ix->d_vectors);
-    if (ix->d_graph) hipFree(ix->d_graph);
-    if (ix->d_tree) hipFree(ix->d_tree);
-    if (ix->d_tree_start) hipFree(ix->d_tree_start);
-    if (ix->d_deleted) hipFree(ix->d_deleted);
+    if (ix->d_vectors) (void)hipFree(ix->d_vectors);
+    if (ix->d_graph) (void)hipFree(ix->d_graph);
+    if (ix->d_tree) (void)hipFree(ix->d_tree);
+    if (ix->d_tree_start) (void)hipFree(ix->d_tree_start);
+    if (ix->d_deleted) (void)hipFree(ix->d_deleted);
+    if (ix->d_visited) (void)hipFree(ix->d_visited);
+    if (ix->d_oflow) (void)hipFree(ix->d_oflow);
+    if (ix->d_stats) (void)hipFree(ix->d_stats);
+    if (ix->d_gng) (void)hipFree(ix->d_gng);
+    if (ix->d_gspt) (void)hipFree(ix->d_gspt);
+    if (ix->ev0) (void)hipEventDestroy(ix->ev0);
+    if (ix->ev1) (void)hipEventDestroy(ix->ev1);
     delete ix;
 }
 
-int sptag_amd_search_batch(SptagAmdIndex* ix, const void* queries, int32_t nq,
-                           int32_t k, int32_t max_check,
-                           int32_t* out_vids, float* out_dists)
+static int ensure_cap(void** p, size_t* cap, size_t need)
 {
-    if (!ix || !queries || nq <= 0 || k <= 0 || k > MAX_K) return SPTAG_AMD_ERR_PARAM;
-    if (!sptag_amd_gpu_available() || !ix->d_vectors) {
-        fprintf(stderr,
-                "sptag_amd: search requires a HIP device (%s); no CPU fallback\n",
-                sptag_amd_build_info());
-        return SPTAG_AMD_ERR_NOGPU;
-    }
-    std::lock_guard<std::mutex> g(ix->lock);
-    HIP_OR_FAIL(hipSetDevice(ix->device), SPTAG_AMD_ERR_NOGPU);
+    if (*cap >= need) return SPTAG_AMD_OK;
+    if (*p) (void)hipFree(*p);
+    *p = nullptr;
+    *cap = 0;
+    if (hipMalloc(p, need) != hipSuccess) return SPTAG_AMD_ERR_OOM;
+    *cap = need;
+    return SPTAG_AMD_OK;
+}
 
+/* core search on DEVICE buffers; scratch cached on the handle. */
+static int search_device_core(SptagAmdIndex* ix, const void* d_q, int32_t nq,
+                              int32_t k, int32_t max_check,
+                              int32_t* d_vids, float* d_dists)
+{
     if (max_check <= 0) max_check = ix->default_maxcheck;
 
     SearchCfg cfg;
@@ -323,131 +341,159 @@ int sptag_amd_search_batch(SptagAmdIndex* ix, const void* queries, int32_t nq,
     cfg.ng_cap = max_check + 4096;
     cfg.spt_cap = 2048;
 
-    /* dynamic-LDS ceiling differs from the 160 KiB hardware LDS on some
-     * runtimes; ask the device and fall back to global heaps when the
-     * budget does not fit. */
     int lds_limit = 64 * 1024;
     (void)hipDeviceGetAttribute(&lds_limit, hipDeviceAttributeMaxSharedMemoryPerBlock,
                                 ix->device);
     bool lds_ok = lds_bytes(ix->dim, ix->esz(), cfg, true) <= (size_t)lds_limit;
 
-    /* device buffers */
+    if (ensure_cap((void**)&ix->d_visited, &ix->visited_cap,
+                   (size_t)nq * cfg.vcap * 4) != SPTAG_AMD_OK) return SPTAG_AMD_ERR_OOM;
+    if (ensure_cap((void**)&ix->d_oflow, &ix->oflow_cap, (size_t)nq * 4) != SPTAG_AMD_OK)
+        return SPTAG_AMD_ERR_OOM;
+    if (ensure_cap((void**)&ix->d_stats, &ix->stats_cap, (size_t)nq * 8) != SPTAG_AMD_OK)
+        return SPTAG_AMD_ERR_OOM;
+    if (!ix->ev0) { (void)hipEventCreate(&ix->ev0); (void)hipEventCreate(&ix->ev1); }
+
+    SearchBufs bufs;
+    bufs.queries = d_q;
+    bufs.out_vids = d_vids;
+    bufs.out_dists = d_dists;
+    bufs.visited = ix->d_visited;
+    bufs.oflow = ix->d_oflow;
+    bufs.stats = ix->d_stats;
+    bufs.gheap_ng = nullptr;
+    bufs.gheap_spt = nullptr;
+
+    ix->last_kernel_ms = 0;
+    bool need_global = !lds_ok;
+    if (lds_ok) {
+        HIP_OR_FAIL(hipMemsetAsync(ix->d_visited, 0, (size_t)nq * cfg.vcap * 4),
+                    SPTAG_AMD_ERR_NOGPU);
+        (void)hipEventRecord(ix->ev0);
+        int err = launch_bkt_search(ix->vt, ix->dm, true, ix->dev(), cfg, bufs, nullptr);
+        if (err != 0) {
+            fprintf(stderr, "sptag_amd: launch failed %d\n", err);
+            return SPTAG_AMD_ERR_INTERNAL;
+        }
+        (void)hipEventRecord(ix->ev1);
+        HIP_OR_FAIL(hipDeviceSynchronize(), SPTAG_AMD_ERR_NOGPU);
+        float ms = 0;
+        (void)hipEventElapsedTime(&ms, ix->ev0, ix->ev1);
+        ix->last_kernel_ms += ms;
+        std::vector<int32_t> oflow(nq);
+        HIP_OR_FAIL(hipMemcpy(oflow.data(), ix->d_oflow, (size_t)nq * 4,
+                              hipMemcpyDeviceToHost), SPTAG_AMD_ERR_NOGPU);
+        for (int32_t i = 0; i < nq; i++)
+            if (oflow[i]) { need_global = true; break; }
+    }
+
+    if (need_global) {
+        /* rerun everything with the reference's own heap capacities
+         * (WorkSpace.h:265) in global scratch — exact reference semantics
+         * even at the full-heap edge. */
+        SearchCfg c2 = cfg;
+        c2.ng_cap = max_check * 30;
+        c2.spt_cap = max_check * 10;
+        if (ensure_cap(&ix->d_gng, &ix->gng_cap,
+                       (size_t)nq * ((size_t)c2.ng_cap + 1) * 8) != SPTAG_AMD_OK)
+            return SPTAG_AMD_ERR_OOM;
+        if (ensure_cap(&ix->d_gspt, &ix->gspt_cap,
+                       (size_t)nq * ((size_t)c2.spt_cap + 1) * 8) != SPTAG_AMD_OK)
+            return SPTAG_AMD_ERR_OOM;
+        SearchBufs b2 = bufs;
+        b2.gheap_ng = ix->d_gng;
+        b2.gheap_spt = ix->d_gspt;
+        HIP_OR_FAIL(hipMemsetAsync(ix->d_visited, 0, (size_t)nq * cfg.vcap * 4),
+                    SPTAG_AMD_ERR_NOGPU);
+        (void)hipEventRecord(ix->ev0);
+        int err = launch_bkt_search(ix->vt, ix->dm, false, ix->dev(), c2, b2, nullptr);
+        if (err != 0) {
+            fprintf(stderr, "sptag_amd: global-variant launch failed %d\n", err);
+            return SPTAG_AMD_ERR_INTERNAL;
+        }
+        (void)hipEventRecord(ix->ev1);
+        HIP_OR_FAIL(hipDeviceSynchronize(), SPTAG_AMD_ERR_NOGPU);
+        float ms = 0;
+        (void)hipEventElapsedTime(&ms, ix->ev0, ix->ev1);
+        ix->last_kernel_ms += ms;
+    }
+
+    std::vector<int32_t> stats(2 * (size_t)nq);
+    HIP_OR_FAIL(hipMemcpy(stats.data(), ix->d_stats, (size_t)nq * 8,
+                          hipMemcpyDeviceToHost), SPTAG_AMD_ERR_NOGPU);
+    ix->last_checked = 0;
+    ix->last_popped = 0;
+    for (int32_t i = 0; i < nq; i++) {
+        ix->last_checked += stats[2 * (size_t)i];
+        ix->last_popped += stats[2 * (size_t)i + 1];
+    }
+    return SPTAG_AMD_OK;
+}
+
+static int check_search_args(SptagAmdIndex* ix, const void* q, int32_t nq, int32_t k)
+{
+    if (!ix || !q || nq <= 0 || k <= 0 || k > MAX_K) return SPTAG_AMD_ERR_PARAM;
+    if (!sptag_amd_gpu_available() || !ix->d_vectors) {
+        fprintf(stderr,
+                "sptag_amd: search requires a HIP device (%s); no CPU fallback\n",
+                sptag_amd_build_info());
+        return SPTAG_AMD_ERR_NOGPU;
+    }
+    return SPTAG_AMD_OK;
+}
+
+int sptag_amd_search_batch_device(SptagAmdIndex* ix, const void* d_queries,
+                                  int32_t nq, int32_t k, int32_t max_check,
+                                  int32_t* d_out_vids, float* d_out_dists)
+{
+    int rc = check_search_args(ix, d_queries, nq, k);
+    if (rc != SPTAG_AMD_OK) return rc;
+    std::lock_guard<std::mutex> g(ix->lock);
+    HIP_OR_FAIL(hipSetDevice(ix->device), SPTAG_AMD_ERR_NOGPU);
+    return search_device_core(ix, d_queries, nq, k, max_check, d_out_vids,
+                              d_out_dists);
+}
+
+int sptag_amd_search_batch(SptagAmdIndex* ix, const void* queries, int32_t nq,
+                           int32_t k, int32_t max_check,
+                           int32_t* out_vids, float* out_dists)
+{
+    int rc = check_search_args(ix, queries, nq, k);
+    if (rc != SPTAG_AMD_OK) return rc;
+    std::lock_guard<std::mutex> g(ix->lock);
+    HIP_OR_FAIL(hipSetDevice(ix->device), SPTAG_AMD_ERR_NOGPU);
+
     size_t qbytes = (size_t)nq * ix->dim * ix->esz();
     void* d_q = nullptr;
     int32_t* d_vids = nullptr;
     float* d_dists = nullptr;
-    int32_t* d_visited = nullptr;
-    int32_t* d_oflow = nullptr;
-    void* d_gng = nullptr;
-    void* d_gspt = nullptr;
-    int rc = SPTAG_AMD_ERR_NOGPU;
-    std::vector<int32_t> oflow(nq);
-
+    rc = SPTAG_AMD_ERR_NOGPU;
     do {
         if (hipMalloc(&d_q, qbytes) != hipSuccess) { rc = SPTAG_AMD_ERR_OOM; break; }
-        if (hipMemcpy(d_q, queries, qbytes, hipMemcpyHostToDevice) != hipSuccess) break;
         if (hipMalloc(&d_vids, (size_t)nq * k * 4) != hipSuccess) { rc = SPTAG_AMD_ERR_OOM; break; }
         if (hipMalloc(&d_dists, (size_t)nq * k * 4) != hipSuccess) { rc = SPTAG_AMD_ERR_OOM; break; }
-        if (hipMalloc(&d_oflow, (size_t)nq * 4) != hipSuccess) { rc = SPTAG_AMD_ERR_OOM; break; }
-        if (hipMalloc(&d_visited, (size_t)nq * cfg.vcap * 4) != hipSuccess) { rc = SPTAG_AMD_ERR_OOM; break; }
-        if (hipMemset(d_visited, 0, (size_t)nq * cfg.vcap * 4) != hipSuccess) break;
-
-        SearchBufs bufs;
-        bufs.queries = d_q;
-        bufs.out_vids = d_vids;
-        bufs.out_dists = d_dists;
-        bufs.visited = d_visited;
-        bufs.oflow = d_oflow;
-        bufs.gheap_ng = nullptr;
-        bufs.gheap_spt = nullptr;
-
-        if (lds_ok) {
-            int err = launch_bkt_search(ix->vt, ix->dm, true, ix->dev(), cfg, bufs, nullptr);
-            if (err != 0) { fprintf(stderr, "sptag_amd: launch failed %d\n", err); break; }
-            if (hipDeviceSynchronize() != hipSuccess) break;
-            if (hipMemcpy(oflow.data(), d_oflow, (size_t)nq * 4, hipMemcpyDeviceToHost)
-                != hipSuccess) break;
-        } else {
-            for (auto& f : oflow) f = 1;   /* go straight to the global variant */
-        }
-
-        /* rerun overflowed queries with the global-heap variant at the
-         * reference's own capacities (WorkSpace.h:265). */
-        std::vector<int32_t> redo;
-        for (int32_t i = 0; i < nq; i++)
-            if (oflow[i]) redo.push_back(i);
-        if (!redo.empty()) {
-            SearchCfg c2 = cfg;
-            c2.nq = (int32_t)redo.size();
-            c2.ng_cap = max_check * 30;
-            c2.spt_cap = max_check * 10;
-            std::vector<char> q2((size_t)redo.size() * ix->dim * ix->esz());
-            for (size_t i = 0; i < redo.size(); i++)
-                memcpy(q2.data() + i * ix->dim * ix->esz(),
-                       (const char*)queries + (size_t)redo[i] * ix->dim * ix->esz(),
-                       ix->dim * ix->esz());
-            void* d_q2 = nullptr;
-            int32_t* d_v2 = nullptr;
-            float* d_d2 = nullptr;
-            if (hipMalloc(&d_q2, q2.size()) != hipSuccess ||
-                hipMalloc(&d_v2, redo.size() * k * 4) != hipSuccess ||
-                hipMalloc(&d_d2, redo.size() * k * 4) != hipSuccess ||
-                hipMalloc(&d_gng, redo.size() * ((size_t)c2.ng_cap + 1) * 8) != hipSuccess ||
-                hipMalloc(&d_gspt, redo.size() * ((size_t)c2.spt_cap + 1) * 8) != hipSuccess) {
-                rc = SPTAG_AMD_ERR_OOM;
-                if (d_q2) hipFree(d_q2);
-                if (d_v2) hipFree(d_v2);
-                if (d_d2) hipFree(d_d2);
-                break;
-            }
-            hipMemcpy(d_q2, q2.data(), q2.size(), hipMemcpyHostToDevice);
-            hipMemset(d_visited, 0, (size_t)redo.size() * cfg.vcap * 4);
-            hipMemset(d_oflow, 0, (size_t)redo.size() * 4);
-            SearchBufs b2 = bufs;
-            b2.queries = d_q2;
-            b2.out_vids = d_v2;
-            b2.out_dists = d_d2;
-            b2.gheap_ng = d_gng;
-            b2.gheap_spt = d_gspt;
-            int err = launch_bkt_search(ix->vt, ix->dm, false, ix->dev(), c2, b2, nullptr);
-            if (err != 0 || hipDeviceSynchronize() != hipSuccess) {
-                fprintf(stderr, "sptag_amd: fallback launch failed %d\n", err);
-                hipFree(d_q2); hipFree(d_v2); hipFree(d_d2);
-                break;
-            }
-            std::vector<int32_t> v2(redo.size() * k);
-            std::vector<float> dd2(redo.size() * k);
-            hipMemcpy(v2.data(), d_v2, v2.size() * 4, hipMemcpyDeviceToHost);
-            hipMemcpy(dd2.data(), d_d2, dd2.size() * 4, hipMemcpyDeviceToHost);
-            hipFree(d_q2); hipFree(d_v2); hipFree(d_d2);
-            /* scatter into the main output after the bulk copy below */
-            if (hipMemcpy(out_vids, d_vids, (size_t)nq * k * 4, hipMemcpyDeviceToHost)
-                != hipSuccess) break;
-            if (hipMemcpy(out_dists, d_dists, (size_t)nq * k * 4, hipMemcpyDeviceToHost)
-                != hipSuccess) break;
-            for (size_t i = 0; i < redo.size(); i++) {
-                memcpy(out_vids + (size_t)redo[i] * k, v2.data() + i * k, (size_t)k * 4);
-                memcpy(out_dists + (size_t)redo[i] * k, dd2.data() + i * k, (size_t)k * 4);
-            }
-            rc = SPTAG_AMD_OK;
-            break;
-        }
-
+        if (hipMemcpy(d_q, queries, qbytes, hipMemcpyHostToDevice) != hipSuccess) break;
+        rc = search_device_core(ix, d_q, nq, k, max_check, d_vids, d_dists);
+        if (rc != SPTAG_AMD_OK) break;
         if (hipMemcpy(out_vids, d_vids, (size_t)nq * k * 4, hipMemcpyDeviceToHost)
-            != hipSuccess) break;
+            != hipSuccess) { rc = SPTAG_AMD_ERR_NOGPU; break; }
         if (hipMemcpy(out_dists, d_dists, (size_t)nq * k * 4, hipMemcpyDeviceToHost)
-            != hipSuccess) break;
+            != hipSuccess) { rc = SPTAG_AMD_ERR_NOGPU; break; }
         rc = SPTAG_AMD_OK;
     } while (0);
-
-    if (d_q) hipFree(d_q);
-    if (d_vids) hipFree(d_vids);
-    if (d_dists) hipFree(d_dists);
-    if (d_visited) hipFree(d_visited);
-    if (d_oflow) hipFree(d_oflow);
-    if (d_gng) hipFree(d_gng);
-    if (d_gspt) hipFree(d_gspt);
+    if (d_q) (void)hipFree(d_q);
+    if (d_vids) (void)hipFree(d_vids);
+    if (d_dists) (void)hipFree(d_dists);
     return rc;
+}
+
+void sptag_amd_last_stats(SptagAmdIndex* ix, double* kernel_ms,
+                          long long* checked, long long* popped)
+{
+    if (!ix) return;
+    if (kernel_ms) *kernel_ms = ix->last_kernel_ms;
+    if (checked) *checked = ix->last_checked;
+    if (popped) *popped = ix->last_popped;
 }
 
 int sptag_amd_truth(SptagAmdIndex* ix, const void* queries, int32_t nq,

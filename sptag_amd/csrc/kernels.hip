@@ -478,12 +478,14 @@ void bkt_search_kernel(DevIndex di, SearchCfg cfg, SearchBufs bufs)
 
     const int deg = di.deg;
     const int checkPos = deg - 1;
+    int popped = 0;
 
     for (;;) {
         __syncthreads();
         if (ss->ng_count <= 0 || ss->terminate || ss->oflow) break;
         if (lane == 0) ss->popped = ndheap_pop(c.ng, &ss->ng_count);
         __syncthreads();
+        popped++;
         NodeDist gnode = ss->popped;
         const int32_t* row = di.graph + (size_t)gnode.node * deg;
         int32_t nn = lane < deg ? row[lane] : -1;
@@ -566,6 +568,10 @@ void bkt_search_kernel(DevIndex di, SearchCfg cfg, SearchBufs bufs)
 
     if (lane == 0) {
         bufs.oflow[q] = ss->oflow;
+        if (bufs.stats) {
+            bufs.stats[(size_t)q * 2 + 0] = ss->checked;
+            bufs.stats[(size_t)q * 2 + 1] = popped;
+        }
         qrs_sort(qrs, cfg.k);
     }
     __syncthreads();
