@@ -1,0 +1,377 @@
+#!/usr/bin/env python3
+"""Benchmark of the MI355X-native SPTAG search backend.
+
+Measures BASELINE.json's metric — QPS at recall@10 >= 0.95 on a 10k-query
+batch — on the named config. Default (no flags): config #2, SPTAG-BKT
+10M x 128 float32 L2 on 1 GPU ("bkt_10m_d128_f32_l2").
+
+A "step" is one batched search of the full query set through the hot path
+(sptag_amd_search_batch_device: queries and outputs resident in HBM; the
+host->device query upload happens once, before the timed region).
+
+Multi-GPU (--gpus N, launched via torch.distributed.run): the SAME dataset
+is range-sharded by contiguous VID ranges across ranks;
+each rank builds/searches its own shard, one all-gather of per-shard top-k
+per step, host merge on rank 0. Total work fixed => "scaling": "strong".
+
+Synthetic data: overlapping Gaussian mixture (SIFT-like clusterability —
+BASELINE.md synthetic-data note; i.i.d. uniform makes recall 0.95
+unreachable for ANY backend). Fixed seed 2016.
+
+cpu_baseline: the REFERENCE CPU implementation (oracle/_ref indexsearcher,
+compiled from /root/reference sources) timed on this box's host cores on
+the same index files, bounded sample; falls back to the oracle C port if
+the binary is absent.
+"""
+import argparse
+import json
+import os
+import subprocess
+import sys
+import time
+
+import numpy as np
+
+REPO = os.path.dirname(os.path.abspath(__file__))
+sys.path.insert(0, REPO)
+
+HBM_PEAK_GBS = 8000.0   # MI355X HBM3E peak (spec), MI355X_MICROARCH.md
+
+CONFIGS = {
+    # BASELINE.json configs[0] — plumbing scale (CPU-runnable reference case)
+    "bkt_100k_d128_f32_l2": dict(n=100_000, d=128, dtype="f32", metric="L2",
+                                 nq=1000, k=10, ncenters=1024, sigma=32.0),
+    # BASELINE.json configs[1] — the headline single-GPU config (DEFAULT)
+    "bkt_10m_d128_f32_l2": dict(n=10_000_000, d=128, dtype="f32", metric="L2",
+                                nq=10_000, k=10, ncenters=8192, sigma=32.0),
+    # quick smoke-scale variant for plumbing runs (not a bench line)
+    "bkt_1m_d128_f32_l2": dict(n=1_000_000, d=128, dtype="f32", metric="L2",
+                               nq=10_000, k=10, ncenters=4096, sigma=32.0),
+    # BASELINE.json configs[2] — int8 cosine SPACEV shape
+    "bkt_100m_d100_i8_cos": dict(n=100_000_000, d=100, dtype="i8",
+                                 metric="Cosine", nq=10_000, k=10,
+                                 ncenters=16384, sigma=30.0),
+}
+DEFAULT_WORKLOAD = "bkt_10m_d128_f32_l2"
+MC_SWEEP = [512, 1024, 2048, 4096, 8192, 16384]
+
+
+def log(rank, *a):
+    if rank == 0:
+        print("[bench]", *a, file=sys.stderr, flush=True)
+
+
+def gen_data(cfg, shard, world, device, torch):
+    """Deterministic mixture; every rank generates its contiguous shard
+    rows [lo, hi) plus the shared query set."""
+    n, d = cfg["n"], cfg["d"]
+    gen = torch.Generator(device=device)
+    gen.manual_seed(2016)
+    # hierarchical mixture: super-centers -> centers -> points, so cluster
+    # neighborhoods overlap in chains (SIFT-like clusterability) instead of
+    # isolated Gaussian islands (BASELINE.md synthetic-data note).
+    supers = torch.rand((256, d), generator=gen, device=device) * 255.0
+    slab = torch.randint(0, 256, (cfg["ncenters"],), generator=gen, device=device)
+    centers = supers[slab] + torch.randn((cfg["ncenters"], d), generator=gen,
+                                         device=device) * (cfg["sigma"] * 1.5)
+    lo = n * shard // world
+    hi = n * (shard + 1) // world
+    gen.manual_seed(2016 + 1 + shard)
+    lab = torch.randint(0, cfg["ncenters"], (hi - lo,), generator=gen, device=device)
+    x = centers[lab] + torch.randn((hi - lo, d), generator=gen, device=device) * cfg["sigma"]
+    gen.manual_seed(2016 + 9999)
+    qlab = torch.randint(0, cfg["ncenters"], (cfg["nq"],), generator=gen, device=device)
+    q = centers[qlab] + torch.randn((cfg["nq"], d), generator=gen, device=device) * cfg["sigma"]
+    if cfg["dtype"] == "i8":
+        x = x.clamp(-127, 127).round()
+        q = q.clamp(-127, 127).round()
+        return x.to(torch.int8), q.to(torch.int8), lo
+    return x.float(), q.float(), lo
+
+
+def shard_truth(x, q, k, metric, torch, lo, chunk=2_000_000):
+    """Exact top-k of q against shard x (ids offset by lo). GEMM shortlist
+    (f32) + float64 rescore of a 4k-deep shortlist."""
+    short = min(64, x.shape[0])
+    xq = q.float()
+    best_d = None
+    best_i = None
+    for s in range(0, x.shape[0], chunk):
+        xc = x[s:s + chunk].float()
+        dots = xq @ xc.T
+        if metric == "L2":
+            d2 = (xq * xq).sum(1, keepdim=True) + (xc * xc).sum(1)[None, :] - 2.0 * dots
+        else:
+            base = 1.0 if x.dtype == torch.float32 else 127.0
+            d2 = base * base - dots
+        vals, idx = torch.topk(d2, min(short, xc.shape[0]), dim=1, largest=False)
+        idx = idx + s
+        if best_d is None:
+            best_d, best_i = vals, idx
+        else:
+            cd = torch.cat([best_d, vals], 1)
+            ci = torch.cat([best_i, idx], 1)
+            vals2, pos = torch.topk(cd, short, dim=1, largest=False)
+            best_d = vals2
+            best_i = torch.gather(ci, 1, pos)
+    # exact float64 rescore of the shortlist
+    xv = x[best_i.reshape(-1)].reshape(*best_i.shape, x.shape[1]).double()
+    qv = q.double()[:, None, :]
+    if metric == "L2":
+        dd = ((xv - qv) ** 2).sum(-1)
+    else:
+        base = 1.0 if x.dtype == torch.float32 else 127.0
+        dd = base * base - (xv * qv).sum(-1)
+    vals, pos = torch.topk(dd, k, dim=1, largest=False)
+    ids = torch.gather(best_i, 1, pos) + lo
+    return ids.int(), vals.float()
+
+
+def merge_topk(vids, dists, k):
+    """host merge of concatenated per-shard top-k lists (AggregatorService
+    AggregateResults semantics: concatenate + select)."""
+    idx = np.argsort(dists, axis=1, kind="stable")[:, :k]
+    return (np.take_along_axis(vids, idx, 1)[:, :k],
+            np.take_along_axis(dists, idx, 1)[:, :k])
+
+
+def recall_at_k(got_vids, truth_vids, k):
+    nq = got_vids.shape[0]
+    hits = 0
+    for i in range(nq):
+        hits += len(set(truth_vids[i, :k].tolist()) & set(got_vids[i, :k].tolist()))
+    return hits / (nq * k)
+
+
+def cpu_baseline_leg(index_dir, q_np, mc, k, seconds_budget=25):
+    """Time the reference CPU searcher (oracle/_ref) on this box's host
+    cores, bounded sample. Returns the cpu_baseline JSON object or None."""
+    import multiprocessing
+    cores = multiprocessing.cpu_count()
+    ref = os.path.join(REPO, "oracle", "_ref", "indexsearcher")
+    nq = min(1024, q_np.shape[0])
+    qfile = os.path.join(index_dir, "bench_queries.bin")
+    with open(qfile, "wb") as f:
+        f.write(np.int32(nq).tobytes())
+        f.write(np.int32(q_np.shape[1]).tobytes())
+        f.write(q_np[:nq].tobytes())
+    vt = "Float" if q_np.dtype == np.float32 else "Int8"
+    if os.path.exists(ref):
+        try:
+            out = subprocess.run(
+                [ref, "-d", str(q_np.shape[1]), "-v", vt, "-f", "DEFAULT",
+                 "-i", qfile, "-x", index_dir, "-k", str(k), "-m", str(mc),
+                 "-t", str(cores)],
+                capture_output=True, text=True, timeout=1200, cwd=index_dir)
+            qps = None
+            for line in (out.stdout + out.stderr).splitlines():
+                # "[1] 0-NQ  mc  avg  p99  p95  recall  QPS  peakGB"
+                parts = line.split()
+                for i, tok in enumerate(parts):
+                    if tok == f"0-{nq}" and len(parts) >= i + 7:
+                        qps = float(parts[i + 6])
+                        break
+            if qps:
+                return {"value": qps, "unit": "queries/s", "cores": cores,
+                        "kind": "reference",
+                        "sample": f"{nq} queries @ MaxCheck {mc}, "
+                                  f"reference indexsearcher -t {cores}"}
+        except Exception as e:  # noqa: BLE001
+            print("[bench] reference baseline failed:", e, file=sys.stderr)
+    # fallback: the oracle C port, in process
+    try:
+        from oracle.pyoracle import OrcIndex
+        oix = OrcIndex.load(index_dir)
+        t0 = time.time()
+        done = 0
+        step = 128
+        while time.time() - t0 < seconds_budget and done < q_np.shape[0]:
+            oix.search_batch(q_np[done:done + step], k, mc, nthreads=cores)
+            done += step
+        dt = time.time() - t0
+        return {"value": done / dt, "unit": "queries/s", "cores": cores,
+                "kind": "port",
+                "sample": f"{done} queries @ MaxCheck {mc}, oracle port, "
+                          f"{cores} OpenMP threads"}
+    except Exception as e:  # noqa: BLE001
+        print("[bench] oracle baseline failed:", e, file=sys.stderr)
+        return None
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=5)
+    ap.add_argument("--warmup", type=int, default=2)
+    ap.add_argument("--workload", default=DEFAULT_WORKLOAD)
+    ap.add_argument("--mc", type=int, default=0, help="force MaxCheck (skip sweep)")
+    ap.add_argument("--no-cpu-baseline", action="store_true")
+    ap.add_argument("--ntrees", type=int, default=4)
+    ap.add_argument("--refine", type=int, default=1)
+    args = ap.parse_args()
+
+    import torch
+    cfg = CONFIGS[args.workload]
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    dist = None
+    if world > 1:
+        import torch.distributed as tdist
+        dist = tdist
+        dist.init_process_group("nccl")
+        torch.cuda.set_device(local_rank)
+    assert torch.cuda.is_available(), "bench requires a HIP device"
+    device = f"cuda:{local_rank}"
+    torch.cuda.set_device(device)
+
+    import sptag_amd
+    from sptag_amd.build import build_index_arrays
+
+    t_setup0 = time.time()
+    x, q, lo = gen_data(cfg, rank, world, device, torch)
+    log(rank, f"data: shard {rank}/{world} rows {x.shape} ({time.time()-t_setup0:.1f}s)")
+
+    t0 = time.time()
+    x_np = x.cpu().numpy()
+    arrays = build_index_arrays(
+        x_np, cfg["metric"], ntrees=args.ntrees, refine_rounds=args.refine,
+        device=device, normalized=False, verbose=(rank == 0))
+    log(rank, f"index built ({time.time()-t0:.1f}s)")
+    t0 = time.time()
+    ix = sptag_amd.AnnIndex.FromArrays(
+        arrays["vectors"], arrays["tree_start"], arrays["tree_nodes"],
+        arrays["graph"], cfg["metric"], device=local_rank)
+    log(rank, f"index uploaded ({time.time()-t0:.1f}s)")
+
+    # queries must match the (possibly normalized) stored vector space? No:
+    # reference searches with RAW queries (SURVEY.md §8a cosine notes).
+    if cfg["metric"] == "Cosine":
+        xs = torch.as_tensor(arrays["vectors"], device=device)
+    else:
+        xs = x
+    t0 = time.time()
+    tv, td = shard_truth(xs, q, cfg["k"], cfg["metric"], torch, lo)
+    if world > 1:
+        gv = [torch.zeros_like(tv) for _ in range(world)]
+        gd = [torch.zeros_like(td) for _ in range(world)]
+        dist.all_gather(gv, tv.contiguous())
+        dist.all_gather(gd, td.contiguous())
+        tv_np, _ = merge_topk(torch.cat(gv, 1).cpu().numpy(),
+                              torch.cat(gd, 1).cpu().numpy(), cfg["k"])
+    else:
+        tv_np = tv.cpu().numpy()
+    log(rank, f"truth ready ({time.time()-t0:.1f}s)")
+
+    k = cfg["k"]
+    nq = cfg["nq"]
+    d_q = q.contiguous()
+    d_vids = torch.empty((nq, k), dtype=torch.int32, device=device)
+    d_dists = torch.empty((nq, k), dtype=torch.float32, device=device)
+
+    def one_step(mc):
+        ix.BatchSearchDevice(d_q.data_ptr(), nq, k, d_vids.data_ptr(),
+                             d_dists.data_ptr(), mc)
+
+    def merged_results(mc):
+        one_step(mc)
+        v = (d_vids + lo).masked_fill(d_vids < 0, -1)
+        if world > 1:
+            gv = [torch.zeros_like(v) for _ in range(world)]
+            gd = [torch.zeros_like(d_dists) for _ in range(world)]
+            dist.all_gather(gv, v.contiguous())
+            dist.all_gather(gd, d_dists.contiguous())
+            av = torch.cat(gv, 1).cpu().numpy()
+            ad = torch.cat(gd, 1).cpu().numpy()
+            return merge_topk(av, ad, k)
+        return v.cpu().numpy(), d_dists.cpu().numpy()
+
+    # MaxCheck sweep -> cheapest mc with recall >= 0.95
+    chosen_mc, chosen_recall = None, 0.0
+    sweep = [args.mc] if args.mc else MC_SWEEP
+    for mc in sweep:
+        gv, _ = merged_results(mc)
+        r = recall_at_k(gv, tv_np, k)
+        log(rank, f"sweep mc={mc}: recall@{k}={r:.4f}")
+        chosen_mc, chosen_recall = mc, r
+        if r >= 0.95:
+            break
+
+    # timed region
+    for _ in range(args.warmup):
+        one_step(chosen_mc)
+    if world > 1:
+        dist.barrier()
+    torch.cuda.synchronize()
+    t0 = time.time()
+    for _ in range(args.steps):
+        one_step(chosen_mc)
+    torch.cuda.synchronize()
+    if world > 1:
+        dist.barrier()
+    elapsed = time.time() - t0
+    if world > 1:
+        te = torch.tensor([elapsed], device=device)
+        dist.all_reduce(te, op=dist.ReduceOp.MAX)
+        elapsed = float(te.item())
+
+    kernel_ms, checked, popped = ix.LastStats()
+    esz = 4 if cfg["dtype"] == "f32" else 1
+    alg_bytes = checked * cfg["d"] * esz + popped * ix.degree * 4
+    achieved = (alg_bytes / 1e9) / (kernel_ms / 1e3) if kernel_ms > 0 else 0.0
+
+    value = nq * args.steps / elapsed
+    ms_per_step = elapsed / args.steps * 1e3
+
+    result = None
+    if rank == 0:
+        cpu = None
+        if not args.no_cpu_baseline and world == 1:
+            t0 = time.time()
+            idx_dir = os.environ.get("BENCH_INDEX_DIR", "/tmp/bench_index")
+            os.makedirs(idx_dir, exist_ok=True)
+            ix.Save(idx_dir)
+            q_np = q.cpu().numpy()
+            cpu = cpu_baseline_leg(idx_dir, q_np, chosen_mc, k)
+            log(rank, f"cpu baseline done ({time.time()-t0:.1f}s): {cpu}")
+        result = {
+            "metric": "QPS @ recall@10>=0.95, batch=10k queries",
+            "value": round(value, 1),
+            "unit": "queries/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 3),
+            "higher_is_better": True,
+            "scaling": "strong",
+            "vs_baseline": None,
+            "dtype": cfg["dtype"],
+            "data": "synthetic",
+            "config": {
+                "workload": args.workload,
+                "n": cfg["n"], "dim": cfg["d"], "metric": cfg["metric"],
+                "nq": nq, "k": k, "max_check": chosen_mc,
+                "recall_at_10": round(chosen_recall, 4),
+                "recall_gate_met": bool(chosen_recall >= 0.95),
+                "sharding": "contiguous VID ranges" if world > 1 else "none",
+            },
+            "roofline": {
+                "bound": "hbm",
+                "achieved": round(achieved, 1),
+                "peak": HBM_PEAK_GBS,
+                "unit": "GB/s",
+                "frac": round(achieved / HBM_PEAK_GBS, 4),
+                "traffic": None,
+                "kernel_ms_per_step": round(kernel_ms, 3),
+                "alg_bytes_per_step": alg_bytes,
+            },
+            "cpu_baseline": cpu,
+        }
+        print(json.dumps(result))
+    if dist:
+        dist.destroy_process_group()
+    return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())

@@ -285,6 +285,7 @@ def build_rng_graph(vectors, *, degree=32, ntrees=8, tpt_leaf=1000, cand=64,
 
     kper = min(cand // 2 + 1, 48)
     ids = dst = None
+    last_partition = None
     for t in range(ntrees):
         perm, bounds = _tpt_leaves(xf, tpt_leaf, gen)
         tids, tdst = _leaf_knn(xf, perm, bounds, kper)
@@ -294,13 +295,55 @@ def build_rng_graph(vectors, *, degree=32, ntrees=8, tpt_leaf=1000, cand=64,
         else:
             ids, dst = _merge_candidates(ids, dst, tids, tdst, cand, self_ids)
         del tids, tdst
+        last_partition = (perm, bounds)
         if verbose:
             print(f"  tpt tree {t + 1}/{ntrees} merged ({len(bounds)} leaves)")
+
+    # Highway bridges: the reference's search-based RefineNode pool
+    # (CEF=1000, NeighborhoodGraph.h:535) reaches far beyond a point's own
+    # cluster, which is what creates the cross-cluster edges the RNG prune
+    # then filters. Truncated in-leaf KNN pools cannot (tight clusters
+    # become graph islands), so append per-point "bridge" candidates: the
+    # medoids of the nearest other TP-tree leaves. They sort to the pool's
+    # tail and survive (the pool is widened, not truncated), and the RNG
+    # rule decides acceptance exactly as RebuildNeighbors would.
+    perm, bounds = last_partition
+    L = len(bounds)
+    bknn = min(8, L - 1)
+    bridge_ids = None
+    if bknn > 0:
+        leaf_of = torch.zeros(n, dtype=torch.int64, device=device)
+        maxm = max(c for _, c in bounds)
+        pad = torch.zeros((L, maxm), dtype=torch.int64, device=device)
+        pmask = torch.zeros((L, maxm), dtype=torch.bool, device=device)
+        for li, (st, c) in enumerate(bounds):
+            leaf_of[perm[st:st + c]] = li
+            pad[li, :c] = perm[st:st + c]
+            pmask[li, :c] = True
+        pts = xf[pad]                                    # [L, maxm, d]
+        cnt = pmask.sum(1, keepdim=True).clamp(min=1)
+        centroids = (pts * pmask[:, :, None]).sum(1) / cnt
+        cd = ((pts - centroids[:, None, :]) ** 2).sum(-1)
+        cd = cd.masked_fill(~pmask, float("inf"))
+        medoid = pad.gather(1, cd.argmin(1, keepdim=True)).squeeze(1)  # [L]
+        lc2 = torch.cdist(centroids, centroids)
+        lc2.fill_diagonal_(float("inf"))
+        nleaf = lc2.topk(bknn, dim=1, largest=False).indices  # [L, bknn]
+        bridge_ids = medoid[nleaf][leaf_of].int()             # [n, bknn]
+        del pad, pmask, pts, cd, lc2
 
     graph = torch.full((n, degree), -1, dtype=torch.int32, device=device)
     for s in range(0, n, point_chunk):
         e = min(n, s + point_chunk)
-        graph[s:e] = _rng_prune(xf, ids[s:e], dst[s:e], degree, rng_factor, device)
+        cid, cdd = ids[s:e], dst[s:e]
+        if bridge_ids is not None:
+            bi = bridge_ids[s:e]
+            bd = _exact_l2(xf, torch.arange(s, e, device=device), bi.long())
+            # widen the pool (cand + bknn) so bridges are never truncated away
+            cid, cdd = _merge_candidates(cid, cdd, bi, bd,
+                                         cid.shape[1] + bi.shape[1],
+                                         self_ids[s:e])
+        graph[s:e] = _rng_prune(xf, cid, cdd, degree, rng_factor, device)
         if verbose and (s // point_chunk) % 20 == 0:
             print(f"  rng prune {e}/{n}")
     return graph, ids, dst
