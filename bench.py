@@ -207,7 +207,7 @@ def main():
     ap.add_argument("--mc", type=int, default=0, help="force MaxCheck (skip sweep)")
     ap.add_argument("--no-cpu-baseline", action="store_true")
     ap.add_argument("--ntrees", type=int, default=4)
-    ap.add_argument("--refine", type=int, default=1)
+    ap.add_argument("--refine", type=int, default=0)
     args = ap.parse_args()
 
     import torch

@@ -267,11 +267,25 @@ def _merge_candidates(ids_a, dst_a, ids_b, dst_b, cand, self_ids):
     invalid = dup | (cid < 0) | (cid == self_ids[:, None])
     cdd = cdd.masked_fill(invalid, float("inf"))
     order = cdd.argsort(dim=1, stable=True)[:, :cand]
-    return torch.gather(cid, 1, order), torch.gather(cdd, 1, order)
+    cid = torch.gather(cid, 1, order)
+    cdd = torch.gather(cdd, 1, order)
+    if cid.shape[1] < cand:
+        padn = cand - cid.shape[1]
+        cid = torch.cat([cid, torch.full((cid.shape[0], padn), -1,
+                                         dtype=cid.dtype, device=cid.device)], 1)
+        cdd = torch.cat([cdd, torch.full((cdd.shape[0], padn), float("inf"),
+                                         device=cdd.device)], 1)
+    cid = cid.masked_fill(~torch.isfinite(cdd), -1)
+    return cid, cdd
 
 
-def build_rng_graph(vectors, *, degree=32, ntrees=8, tpt_leaf=1000, cand=64,
-                    rng_factor=1.0, seed=2016, device=None, point_chunk=100_000,
+def _prune_chunk(cand, d):
+    """point-chunk size keeping the [B, cand, cand] pairwise tensor ~2GB"""
+    return max(4096, int(2e9 // (cand * cand * 4 + cand * d * 8)))
+
+
+def build_rng_graph(vectors, *, degree=32, ntrees=4, tpt_leaf=1000, cand=256,
+                    rng_factor=1.0, seed=2016, device=None, point_chunk=None,
                     verbose=False):
     """Returns graph int32 [n, degree] (RNG-pruned, ascending, -1 padded)."""
     device = device or _dev()
@@ -283,7 +297,7 @@ def build_rng_graph(vectors, *, degree=32, ntrees=8, tpt_leaf=1000, cand=64,
     gen.manual_seed(seed + 77)
     self_ids = torch.arange(n, device=device, dtype=torch.int32)
 
-    kper = min(cand // 2 + 1, 48)
+    kper = min(max(33, cand // max(ntrees, 1)), 64)
     ids = dst = None
     last_partition = None
     for t in range(ntrees):
@@ -332,6 +346,8 @@ def build_rng_graph(vectors, *, degree=32, ntrees=8, tpt_leaf=1000, cand=64,
         bridge_ids = medoid[nleaf][leaf_of].int()             # [n, bknn]
         del pad, pmask, pts, cd, lc2
 
+    if point_chunk is None:
+        point_chunk = _prune_chunk(cand, xf.shape[1])
     graph = torch.full((n, degree), -1, dtype=torch.int32, device=device)
     for s in range(0, n, point_chunk):
         e = min(n, s + point_chunk)
@@ -370,9 +386,9 @@ def _rng_prune(xf, cid, cdd, degree, rng_factor, device):
     return out
 
 
-def refine_graph(vectors, graph, cand_ids, cand_dst, *, degree=32, cand=64,
+def refine_graph(vectors, graph, cand_ids, cand_dst, *, degree=32, cand=256,
                  rounds=2, hop_sample=8, rng_factor=1.0, device=None,
-                 point_chunk=100_000, seed=2016, verbose=False):
+                 point_chunk=None, seed=2016, verbose=False):
     """Neighborhood refinement: the reference refines each node's edges from
     a CEF-sized candidate pool gathered by searching the index itself
     (NeighborhoodGraph.h:460-560 RefineGraph/RefineNode). Here the pool is
@@ -388,6 +404,8 @@ def refine_graph(vectors, graph, cand_ids, cand_dst, *, degree=32, cand=64,
     self_ids = torch.arange(n, device=device, dtype=torch.int32)
     gen = torch.Generator(device=device)
     gen.manual_seed(seed + 123)
+    if point_chunk is None:
+        point_chunk = _prune_chunk(cand, xf.shape[1])
 
     for r in range(rounds):
         # reverse edges (sampled): every edge (i -> j) proposes i to j
@@ -434,9 +452,9 @@ def _exact_l2(xf, q_rows, c_ids):
     return d.masked_fill(c_ids < 0, float("inf"))
 
 
-def build_index_arrays(vectors, distmethod, *, degree=32, ntrees=8,
-                       tpt_leaf=1000, cand=64, kmeans_k=32, leaf_size=32,
-                       refine_rounds=2, seed=2016, device=None,
+def build_index_arrays(vectors, distmethod, *, degree=32, ntrees=4,
+                       tpt_leaf=1000, cand=256, kmeans_k=32, leaf_size=32,
+                       refine_rounds=0, seed=2016, device=None,
                        normalized=False, verbose=False):
     """Full build: returns dict(vectors, tree_start, tree_nodes, graph) ready
     for AnnIndex.FromArrays (vectors already cosine-normalized when needed,
