@@ -89,10 +89,12 @@ def gen_data(cfg, shard, world, device, torch):
     return x.float(), q.float(), lo
 
 
-def shard_truth(x, q, k, metric, torch, lo, chunk=2_000_000):
+def shard_truth(x, q, k, metric, torch, lo, chunk=None):
     """Exact top-k of q against shard x (ids offset by lo). GEMM shortlist
-    (f32) + float64 rescore of a 4k-deep shortlist."""
+    (f32) + float64 rescore of a 64-deep shortlist."""
     short = min(64, x.shape[0])
+    if chunk is None:   # keep the [nq, chunk] distance tile ~2GB
+        chunk = max(65_536, int(2e9 // (max(q.shape[0], 1) * 4)))
     xq = q.float()
     best_d = None
     best_i = None
@@ -244,12 +246,14 @@ def main():
         arrays["graph"], cfg["metric"], device=local_rank)
     log(rank, f"index uploaded ({time.time()-t0:.1f}s)")
 
-    # queries must match the (possibly normalized) stored vector space? No:
-    # reference searches with RAW queries (SURVEY.md §8a cosine notes).
+    # truth runs in the STORED vector space (cosine bases are normalized on
+    # disk; queries stay RAW — SURVEY.md §8a cosine notes).
     if cfg["metric"] == "Cosine":
         xs = torch.as_tensor(arrays["vectors"], device=device)
     else:
         xs = x
+    del arrays
+    torch.cuda.empty_cache()
     t0 = time.time()
     tv, td = shard_truth(xs, q, cfg["k"], cfg["metric"], torch, lo)
     if world > 1:
