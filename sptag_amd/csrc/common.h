@@ -69,6 +69,10 @@ inline size_t lds_bytes(int dim, size_t esz, const SearchCfg& c, bool heaps_in_l
 int launch_bkt_search(int valuetype, int distmethod, bool heaps_in_lds,
                       const DevIndex& di, const SearchCfg& cfg,
                       const SearchBufs& bufs, void* stream);
+int launch_gather_rows(void* dst, const void* src, int row_bytes,
+                       const int32_t* d_idx, int nrows, void* stream);
+int launch_scatter_rows(void* dst, const void* src, int row_bytes,
+                        const int32_t* d_idx, int nrows, void* stream);
 int launch_truth(int valuetype, int distmethod, const DevIndex& di,
                  const void* queries, int32_t nq, int32_t k,
                  int32_t* out_vids, float* out_dists, void* stream);

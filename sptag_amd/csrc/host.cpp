@@ -56,6 +56,7 @@ struct HostIndex {
     int32_t* d_stats = nullptr;   size_t stats_cap = 0;
     void* d_gng = nullptr;        size_t gng_cap = 0;
     void* d_gspt = nullptr;       size_t gspt_cap = 0;
+    void* d_redo = nullptr;       size_t redo_cap = 0;
     hipEvent_t ev0 = nullptr, ev1 = nullptr;
     double last_kernel_ms = 0;
     long long last_checked = 0, last_popped = 0;
@@ -305,6 +306,7 @@ void sptag_amd_free_index(SptagAmdIndex* ix)
     if (ix->d_stats) (void)hipFree(ix->d_stats);
     if (ix->d_gng) (void)hipFree(ix->d_gng);
     if (ix->d_gspt) (void)hipFree(ix->d_gspt);
+    if (ix->d_redo) (void)hipFree(ix->d_redo);
     if (ix->ev0) (void)hipEventDestroy(ix->ev0);
     if (ix->ev1) (void)hipEventDestroy(ix->ev1);
     delete ix;
@@ -335,11 +337,13 @@ static int search_device_core(SptagAmdIndex* ix, const void* d_q, int32_t nq,
     cfg.init_pivots = ix->init_pivots;
     cfg.other_pivots = ix->other_pivots;
     cfg.dpq_cap = std::max(max_check / 16, k);   /* WorkSpace.h:268 */
-    cfg.vcap = (int32_t)next_pow2((uint32_t)std::max(4096, max_check * 6));
-    /* LDS-variant capacities: generous slack over the expected occupancy
-     * (~checked + tree inserts); overflow reruns use reference capacities. */
-    cfg.ng_cap = max_check + 4096;
-    cfg.spt_cap = 2048;
+    cfg.vcap = (int32_t)next_pow2((uint32_t)std::max(4096, max_check * 4));
+    /* LDS-variant capacities: sized to the TYPICAL traversal occupancy so
+     * several workgroups fit per CU; the rare query that outgrows them is
+     * rerun on the global-heap variant at the reference's own capacities
+     * (WorkSpace.h:265) — a speed tradeoff, never a semantic one. */
+    cfg.ng_cap = max_check + 1024;
+    cfg.spt_cap = 1024;
 
     int lds_limit = 64 * 1024;
     (void)hipDeviceGetAttribute(&lds_limit, hipDeviceAttributeMaxSharedMemoryPerBlock,
@@ -365,7 +369,11 @@ static int search_device_core(SptagAmdIndex* ix, const void* d_q, int32_t nq,
     bufs.gheap_spt = nullptr;
 
     ix->last_kernel_ms = 0;
-    bool need_global = !lds_ok;
+    ix->last_checked = 0;
+    ix->last_popped = 0;
+    std::vector<int32_t> stats(2 * (size_t)nq);
+    std::vector<int32_t> redo;
+    bool all_global = !lds_ok;
     if (lds_ok) {
         HIP_OR_FAIL(hipMemsetAsync(ix->d_visited, 0, (size_t)nq * cfg.vcap * 4),
                     SPTAG_AMD_ERR_NOGPU);
@@ -383,27 +391,58 @@ static int search_device_core(SptagAmdIndex* ix, const void* d_q, int32_t nq,
         std::vector<int32_t> oflow(nq);
         HIP_OR_FAIL(hipMemcpy(oflow.data(), ix->d_oflow, (size_t)nq * 4,
                               hipMemcpyDeviceToHost), SPTAG_AMD_ERR_NOGPU);
-        for (int32_t i = 0; i < nq; i++)
-            if (oflow[i]) { need_global = true; break; }
+        HIP_OR_FAIL(hipMemcpy(stats.data(), ix->d_stats, (size_t)nq * 8,
+                              hipMemcpyDeviceToHost), SPTAG_AMD_ERR_NOGPU);
+        for (int32_t i = 0; i < nq; i++) {
+            if (oflow[i]) {
+                redo.push_back(i);
+            } else {
+                ix->last_checked += stats[2 * (size_t)i];
+                ix->last_popped += stats[2 * (size_t)i + 1];
+            }
+        }
+        if (redo.size() * 2 > (size_t)nq) {
+            all_global = true;   /* cheaper to redo the whole batch */
+            redo.clear();
+            ix->last_checked = 0;
+            ix->last_popped = 0;
+        }
     }
 
-    if (need_global) {
-        /* rerun everything with the reference's own heap capacities
-         * (WorkSpace.h:265) in global scratch — exact reference semantics
-         * even at the full-heap edge. */
+    int32_t gq_n = all_global ? nq : (int32_t)redo.size();
+    if (gq_n > 0) {
         SearchCfg c2 = cfg;
-        c2.ng_cap = max_check * 30;
+        c2.nq = gq_n;
+        c2.ng_cap = max_check * 30;     /* reference capacities */
         c2.spt_cap = max_check * 10;
         if (ensure_cap(&ix->d_gng, &ix->gng_cap,
-                       (size_t)nq * ((size_t)c2.ng_cap + 1) * 8) != SPTAG_AMD_OK)
+                       (size_t)gq_n * ((size_t)c2.ng_cap + 1) * 8) != SPTAG_AMD_OK)
             return SPTAG_AMD_ERR_OOM;
         if (ensure_cap(&ix->d_gspt, &ix->gspt_cap,
-                       (size_t)nq * ((size_t)c2.spt_cap + 1) * 8) != SPTAG_AMD_OK)
+                       (size_t)gq_n * ((size_t)c2.spt_cap + 1) * 8) != SPTAG_AMD_OK)
             return SPTAG_AMD_ERR_OOM;
         SearchBufs b2 = bufs;
         b2.gheap_ng = ix->d_gng;
         b2.gheap_spt = ix->d_gspt;
-        HIP_OR_FAIL(hipMemsetAsync(ix->d_visited, 0, (size_t)nq * cfg.vcap * 4),
+        size_t row = (size_t)ix->dim * ix->esz();
+        if (!all_global) {
+            /* compact the flagged queries on-device */
+            size_t qseg = ((size_t)gq_n * row + 15) & ~15ul;  /* align */
+            size_t need = (size_t)gq_n * 4 + 16 + qseg + (size_t)gq_n * k * 8;
+            if (ensure_cap(&ix->d_redo, &ix->redo_cap, need) != SPTAG_AMD_OK)
+                return SPTAG_AMD_ERR_OOM;
+            int32_t* d_idx = (int32_t*)ix->d_redo;
+            char* d_q2 = (char*)ix->d_redo + (((size_t)gq_n * 4 + 15) & ~15ul);
+            int32_t* d_v2 = (int32_t*)(d_q2 + qseg);
+            float* d_d2 = (float*)(d_v2 + (size_t)gq_n * k);
+            HIP_OR_FAIL(hipMemcpy(d_idx, redo.data(), (size_t)gq_n * 4,
+                                  hipMemcpyHostToDevice), SPTAG_AMD_ERR_NOGPU);
+            launch_gather_rows(d_q2, d_q, (int)row, d_idx, gq_n, nullptr);
+            b2.queries = d_q2;
+            b2.out_vids = d_v2;
+            b2.out_dists = d_d2;
+        }
+        HIP_OR_FAIL(hipMemsetAsync(ix->d_visited, 0, (size_t)gq_n * cfg.vcap * 4),
                     SPTAG_AMD_ERR_NOGPU);
         (void)hipEventRecord(ix->ev0);
         int err = launch_bkt_search(ix->vt, ix->dm, false, ix->dev(), c2, b2, nullptr);
@@ -412,20 +451,21 @@ static int search_device_core(SptagAmdIndex* ix, const void* d_q, int32_t nq,
             return SPTAG_AMD_ERR_INTERNAL;
         }
         (void)hipEventRecord(ix->ev1);
+        if (!all_global) {
+            int32_t* d_idx = (int32_t*)ix->d_redo;
+            launch_scatter_rows(d_vids, b2.out_vids, k * 4, d_idx, gq_n, nullptr);
+            launch_scatter_rows(d_dists, b2.out_dists, k * 4, d_idx, gq_n, nullptr);
+        }
         HIP_OR_FAIL(hipDeviceSynchronize(), SPTAG_AMD_ERR_NOGPU);
         float ms = 0;
         (void)hipEventElapsedTime(&ms, ix->ev0, ix->ev1);
         ix->last_kernel_ms += ms;
-    }
-
-    std::vector<int32_t> stats(2 * (size_t)nq);
-    HIP_OR_FAIL(hipMemcpy(stats.data(), ix->d_stats, (size_t)nq * 8,
-                          hipMemcpyDeviceToHost), SPTAG_AMD_ERR_NOGPU);
-    ix->last_checked = 0;
-    ix->last_popped = 0;
-    for (int32_t i = 0; i < nq; i++) {
-        ix->last_checked += stats[2 * (size_t)i];
-        ix->last_popped += stats[2 * (size_t)i + 1];
+        HIP_OR_FAIL(hipMemcpy(stats.data(), ix->d_stats, (size_t)gq_n * 8,
+                              hipMemcpyDeviceToHost), SPTAG_AMD_ERR_NOGPU);
+        for (int32_t i = 0; i < gq_n; i++) {
+            ix->last_checked += stats[2 * (size_t)i];
+            ix->last_popped += stats[2 * (size_t)i + 1];
+        }
     }
     return SPTAG_AMD_OK;
 }

@@ -206,11 +206,45 @@ DEV int visited_test_insert(int32_t* tab, uint32_t mask, int32_t idx, int* oflow
 /* float: the AVX512 chunk/fold order of DistanceUtils.cpp:650 (L2) /
  * the cosine analog, with fused lane accumulate (fmaf) as compiled in
  * the reference build (see oracle/sptag_oracle.c header). Result valid
- * on every lane of the group. */
+ * on every lane of the group.
+ *
+ * The fixed-dim form fully unrolls the 16-element chunk loop so the
+ * per-candidate global loads all issue before the fma chain waits on
+ * them (the rolled loop costs one memory latency PER CHUNK — 8x the
+ * stalls at d=128). Dims must be multiples of 16. */
+template <int DM, int DFIX>
+DEV float dist_f32_fix(const float* __restrict__ q, const float* __restrict__ v)
+{
+    const int g = threadIdx.x & 15;
+    constexpr int C = DFIX / 16;
+    float x[C], y[C];
+#pragma unroll
+    for (int c = 0; c < C; c++) { x[c] = q[c * 16 + g]; y[c] = v[c * 16 + g]; }
+    float a = 0.0f;
+#pragma unroll
+    for (int c = 0; c < C; c++)
+        a = (DM == DM_L2) ? fmaf(x[c] - y[c], x[c] - y[c], a) : fmaf(x[c], y[c], a);
+    a = a + __shfl_down(a, 8, 16);
+    a = a + __shfl_down(a, 4, 16);
+    float sv = ((__shfl(a, 0, 16) + __shfl(a, 1, 16)) + __shfl(a, 2, 16)) + __shfl(a, 3, 16);
+    return (DM == DM_L2) ? sv : 1.0f - sv;
+}
+
 template <int DM>
 DEV float ref_dist_grp_f32(const float* __restrict__ q,
                            const float* __restrict__ v, int d)
 {
+    switch (d) {
+    case 32:  return dist_f32_fix<DM, 32>(q, v);
+    case 48:  return dist_f32_fix<DM, 48>(q, v);
+    case 64:  return dist_f32_fix<DM, 64>(q, v);
+    case 96:  return dist_f32_fix<DM, 96>(q, v);
+    case 128: return dist_f32_fix<DM, 128>(q, v);
+    case 256: return dist_f32_fix<DM, 256>(q, v);
+    case 512: return dist_f32_fix<DM, 512>(q, v);
+    case 768: return dist_f32_fix<DM, 768>(q, v);
+    default: break;
+    }
     const int g = threadIdx.x & 15;
     float a = 0.0f;
     const int nd16 = (d >> 4) << 4;
@@ -631,6 +665,60 @@ void truth_kernel(DevIndex di, const void* queries, int32_t nq, int32_t k,
         out_vids[(size_t)q * k + lane] = qrs[lane].vid;
         out_dists[(size_t)q * k + lane] = qrs[lane].dist;
     }
+}
+
+/* ------------------------------------------------------------------ *
+ * row gather/scatter (flagged-query reruns)
+ * ------------------------------------------------------------------ */
+
+__global__ void gather_rows_kernel(char* dst, const char* src, int row_bytes,
+                                   const int32_t* idx, int nrows)
+{
+    int r = blockIdx.x;
+    if (r >= nrows) return;
+    const char* s = src + (size_t)idx[r] * row_bytes;
+    char* d = dst + (size_t)r * row_bytes;
+    if ((row_bytes & 3) == 0) {
+        for (int i = threadIdx.x; i < row_bytes / 4; i += blockDim.x)
+            ((uint32_t*)d)[i] = ((const uint32_t*)s)[i];
+    } else {
+        for (int i = threadIdx.x; i < row_bytes; i += blockDim.x)
+            d[i] = s[i];
+    }
+}
+
+__global__ void scatter_rows_kernel(char* dst, const char* src, int row_bytes,
+                                    const int32_t* idx, int nrows)
+{
+    int r = blockIdx.x;
+    if (r >= nrows) return;
+    const char* s = src + (size_t)r * row_bytes;
+    char* d = dst + (size_t)idx[r] * row_bytes;
+    if ((row_bytes & 3) == 0) {
+        for (int i = threadIdx.x; i < row_bytes / 4; i += blockDim.x)
+            ((uint32_t*)d)[i] = ((const uint32_t*)s)[i];
+    } else {
+        for (int i = threadIdx.x; i < row_bytes; i += blockDim.x)
+            d[i] = s[i];
+    }
+}
+
+int launch_gather_rows(void* dst, const void* src, int row_bytes,
+                       const int32_t* d_idx, int nrows, void* stream)
+{
+    hipLaunchKernelGGL(gather_rows_kernel, dim3(nrows), dim3(64), 0,
+                       (hipStream_t)stream, (char*)dst, (const char*)src,
+                       row_bytes, d_idx, nrows);
+    return (int)hipGetLastError();
+}
+
+int launch_scatter_rows(void* dst, const void* src, int row_bytes,
+                        const int32_t* d_idx, int nrows, void* stream)
+{
+    hipLaunchKernelGGL(scatter_rows_kernel, dim3(nrows), dim3(64), 0,
+                       (hipStream_t)stream, (char*)dst, (const char*)src,
+                       row_bytes, d_idx, nrows);
+    return (int)hipGetLastError();
 }
 
 /* ------------------------------------------------------------------ *
