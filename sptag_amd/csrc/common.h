@@ -59,8 +59,7 @@ inline size_t lds_bytes(int dim, size_t esz, const SearchCfg& c, bool heaps_in_l
     b += ((size_t)c.dpq_cap + 1) * 4;                 /* DistPriorityQueue */
     b += 64;                                          /* scalar slots, padding */
     if (heaps_in_lds) {
-        b += ((size_t)c.ng_cap + 1) * 8;
-        b += ((size_t)c.spt_cap + 1) * 8;
+        b += ((size_t)c.ng_cap + 1) * 8;   /* SPT heap is always global */
     }
     return b;
 }

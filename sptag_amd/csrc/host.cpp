@@ -57,6 +57,7 @@ struct HostIndex {
     void* d_gng = nullptr;        size_t gng_cap = 0;
     void* d_gspt = nullptr;       size_t gspt_cap = 0;
     void* d_redo = nullptr;       size_t redo_cap = 0;
+    void* d_gspt0 = nullptr;      size_t gspt0_cap = 0;
     hipEvent_t ev0 = nullptr, ev1 = nullptr;
     double last_kernel_ms = 0;
     long long last_checked = 0, last_popped = 0;
@@ -307,6 +308,7 @@ void sptag_amd_free_index(SptagAmdIndex* ix)
     if (ix->d_gng) (void)hipFree(ix->d_gng);
     if (ix->d_gspt) (void)hipFree(ix->d_gspt);
     if (ix->d_redo) (void)hipFree(ix->d_redo);
+    if (ix->d_gspt0) (void)hipFree(ix->d_gspt0);
     if (ix->ev0) (void)hipEventDestroy(ix->ev0);
     if (ix->ev1) (void)hipEventDestroy(ix->ev1);
     delete ix;
@@ -342,8 +344,8 @@ static int search_device_core(SptagAmdIndex* ix, const void* d_q, int32_t nq,
      * several workgroups fit per CU; the rare query that outgrows them is
      * rerun on the global-heap variant at the reference's own capacities
      * (WorkSpace.h:265) — a speed tradeoff, never a semantic one. */
-    cfg.ng_cap = max_check + 1024;
-    cfg.spt_cap = 1024;
+    cfg.ng_cap = max_check / 2 + 512;
+    cfg.spt_cap = 4096;
 
     int lds_limit = 64 * 1024;
     (void)hipDeviceGetAttribute(&lds_limit, hipDeviceAttributeMaxSharedMemoryPerBlock,
@@ -358,6 +360,10 @@ static int search_device_core(SptagAmdIndex* ix, const void* d_q, int32_t nq,
         return SPTAG_AMD_ERR_OOM;
     if (!ix->ev0) { (void)hipEventCreate(&ix->ev0); (void)hipEventCreate(&ix->ev1); }
 
+    if (ensure_cap(&ix->d_gspt0, &ix->gspt0_cap,
+                   (size_t)nq * ((size_t)cfg.spt_cap + 1) * 8) != SPTAG_AMD_OK)
+        return SPTAG_AMD_ERR_OOM;
+
     SearchBufs bufs;
     bufs.queries = d_q;
     bufs.out_vids = d_vids;
@@ -366,7 +372,7 @@ static int search_device_core(SptagAmdIndex* ix, const void* d_q, int32_t nq,
     bufs.oflow = ix->d_oflow;
     bufs.stats = ix->d_stats;
     bufs.gheap_ng = nullptr;
-    bufs.gheap_spt = nullptr;
+    bufs.gheap_spt = ix->d_gspt0;
 
     ix->last_kernel_ms = 0;
     ix->last_checked = 0;

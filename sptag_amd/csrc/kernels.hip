@@ -480,12 +480,14 @@ void bkt_search_kernel(DevIndex di, SearchCfg cfg, SearchBufs bufs)
     HeapRef ng, spt;
     ng.cap = cfg.ng_cap;  ng.ref_cap = cfg.max_check * 30;   /* WorkSpace.h:265 */
     spt.cap = cfg.spt_cap; spt.ref_cap = cfg.max_check * 10;
+    /* SPT (tree) heap always lives in global scratch: it is touched only in
+     * the short seed/re-seed phases, and freeing its LDS doubles resident
+     * queries per CU. */
+    spt.a = (NodeDist*)bufs.gheap_spt + (size_t)q * (cfg.spt_cap + 1);
     if (LDSHEAP) {
         ng.a = (NodeDist*)(smem + off); off += ((size_t)cfg.ng_cap + 1) * 8;
-        spt.a = (NodeDist*)(smem + off); off += ((size_t)cfg.spt_cap + 1) * 8;
     } else {
         ng.a = (NodeDist*)bufs.gheap_ng + (size_t)q * (cfg.ng_cap + 1);
-        spt.a = (NodeDist*)bufs.gheap_spt + (size_t)q * (cfg.spt_cap + 1);
     }
 
     /* stage query */
@@ -597,6 +599,18 @@ void bkt_search_kernel(DevIndex di, SearchCfg cfg, SearchBufs bufs)
         __syncthreads();
         if (ss->want_tree)
             search_trees_dev<T, DM>(c, cfg.other_pivots + ss->checked);
+        /* warm L1/L2 with the next pop's adjacency row (speculative,
+         * perf-only: the top rarely changes before the next iteration). */
+        {
+            int32_t nxt = ss->popped.node;   /* broadcast slot reused below */
+            if (lane == 0) ss->popped = ndheap_top(c.ng, ss->ng_count);
+            __syncthreads();
+            nxt = ss->popped.node;
+            if (nxt >= 0 && lane < deg) {
+                int32_t t = di.graph[(size_t)nxt * deg + lane];
+                asm volatile("" :: "v"(t));
+            }
+        }
     }
     __syncthreads();
 
