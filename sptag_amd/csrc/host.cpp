@@ -344,7 +344,7 @@ static int search_device_core(SptagAmdIndex* ix, const void* d_q, int32_t nq,
      * several workgroups fit per CU; the rare query that outgrows them is
      * rerun on the global-heap variant at the reference's own capacities
      * (WorkSpace.h:265) — a speed tradeoff, never a semantic one. */
-    cfg.ng_cap = max_check / 2 + 512;
+    cfg.ng_cap = max_check / 2;
     cfg.spt_cap = 4096;
 
     int lds_limit = 64 * 1024;

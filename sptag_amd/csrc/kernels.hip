@@ -310,6 +310,33 @@ DEV float ref_dist_grp_i8(const int8_t* __restrict__ q,
     return (DM == DM_L2) ? (float)s : (float)(16129 - s);
 }
 
+/* load/compute split of the fixed-dim distance, for software-pipelining
+ * candidate rounds (loads of round r+1 issue while round r reduces). */
+template <int DM, int DFIX>
+struct FragF32 {
+    static constexpr int C = DFIX / 16;
+    float x[C], y[C];
+    DEV void load(const float* __restrict__ q, const float* __restrict__ v)
+    {
+        const int g = threadIdx.x & 15;
+#pragma unroll
+        for (int c = 0; c < C; c++) { x[c] = q[c * 16 + g]; y[c] = v[c * 16 + g]; }
+    }
+    DEV float reduce() const
+    {
+        float a = 0.0f;
+#pragma unroll
+        for (int c = 0; c < C; c++)
+            a = (DM == DM_L2) ? fmaf(x[c] - y[c], x[c] - y[c], a)
+                              : fmaf(x[c], y[c], a);
+        a = a + __shfl_down(a, 8, 16);
+        a = a + __shfl_down(a, 4, 16);
+        float sv = ((__shfl(a, 0, 16) + __shfl(a, 1, 16)) + __shfl(a, 2, 16)) +
+                   __shfl(a, 3, 16);
+        return (DM == DM_L2) ? sv : 1.0f - sv;
+    }
+};
+
 template <typename T, int DM>
 DEV float ref_dist_grp(const T* q, const T* v, int d);
 template <> DEV float ref_dist_grp<float, DM_L2>(const float* q, const float* v, int d)
@@ -361,11 +388,45 @@ DEV int not_deleted(const DevIndex& di, int32_t v)
     return di.deleted[v] == 0;
 }
 
+/* pipelined fixed-dim staging: group g handles candidates g, g+4, ...,
+ * double-buffering fragments so the next round's global loads are in
+ * flight while the current round reduces. */
+template <int DM, int DFIX>
+DEV void stage_dists_pipe_f32(QCtx<float>& c, int cnt)
+{
+    const DevIndex& di = *c.di;
+    const int grp = c.lane >> 4;
+    FragF32<DM, DFIX> A, B;
+    int j = grp;
+    if (j < cnt) A.load(c.qlds, vec_at<float>(di, c.istage[j]));
+    while (j < cnt) {
+        int jn = j + 4;
+        if (jn < cnt) B.load(c.qlds, vec_at<float>(di, c.istage[jn]));
+        float dv = A.reduce();
+        if ((c.lane & 15) == 0) c.dstage[j] = dv;
+        A = B;
+        j = jn;
+    }
+    __syncthreads();
+}
+
 /* stage distances of `cnt` (<=64) data vectors whose ids are in istage[0..cnt)
- * into dstage[0..cnt); whole-wave cooperative, 4 at a time. */
+ * into dstage[0..cnt); whole-wave cooperative, 4 candidates in flight. */
 template <typename T, int DM>
 DEV void stage_dists(QCtx<T>& c, int cnt)
 {
+    if constexpr (sizeof(T) == 4) {
+        QCtx<float>& cf = reinterpret_cast<QCtx<float>&>(c);
+        switch (c.di->dim) {
+        case 32:  stage_dists_pipe_f32<DM, 32>(cf, cnt); return;
+        case 48:  stage_dists_pipe_f32<DM, 48>(cf, cnt); return;
+        case 64:  stage_dists_pipe_f32<DM, 64>(cf, cnt); return;
+        case 96:  stage_dists_pipe_f32<DM, 96>(cf, cnt); return;
+        case 128: stage_dists_pipe_f32<DM, 128>(cf, cnt); return;
+        case 256: stage_dists_pipe_f32<DM, 256>(cf, cnt); return;
+        default: break;
+        }
+    }
     for (int r = 0; r < cnt; r += 4) {
         int j = r + (c.lane >> 4);
         if (j < cnt) {
@@ -558,39 +619,25 @@ void bkt_search_kernel(DevIndex di, SearchCfg cfg, SearchBufs bufs)
         __syncthreads();
         if (ss->terminate) break;
 
-        /* neighbor expansion (BKTIndex.cpp:333-345) */
+        /* neighbor expansion (BKTIndex.cpp:333-345): compact the unvisited
+         * neighbors (row order preserved) and stage their distances. */
         int already = 1;
         if (lane < firstneg)
             already = visited_test_insert(c.vtab, c.vmask, nn, &ss->oflow);
         uint64_t candm = __ballot(lane < firstneg && !already);
-        istage[lane] = nn;
         int ncand = __popcll(candm);
         if ((candm >> lane) & 1) {
             int pos = __popcll(candm & ((1ull << lane) - 1));
-            dstage[pos] = __int_as_float(lane);   /* compacted lane list */
+            istage[pos] = nn;
         }
         __syncthreads();
-        /* move compacted lane ids out of dstage before distances overwrite */
-        int clane = lane < ncand ? __float_as_int(dstage[lane]) : -1;
-        int32_t cvid = clane >= 0 ? istage[clane] : -1;
-        __syncthreads();
-        for (int r = 0; r < ncand; r += 4) {
-            int j = r + (lane >> 4);
-            int32_t vid = __shfl(cvid, j & 63);   /* candidate j's vid */
-            int32_t cl = __shfl(clane, j & 63);
-            if (j < ncand) {
-                float dv = ref_dist_grp<T, DM>(qlds, vec_at<T>(di, vid), di.dim);
-                if ((lane & 15) == 0) dstage[cl] = dv;
-            }
-        }
-        __syncthreads();
+        stage_dists<T, DM>(c, ncand);
         if (lane == 0) {
-            for (int i = 0; i < firstneg; i++) {
-                if (!((candm >> i) & 1)) continue;
-                float dv = dstage[i];
+            for (int r = 0; r < ncand; r++) {
+                float dv = dstage[r];
                 ss->checked++;
                 if (dpq_insert(dpq, &ss->dpq_len, cfg.dpq_cap, dv))
-                    ndheap_insert(c.ng, &ss->ng_count, NodeDist{istage[i], dv}, &ss->oflow);
+                    ndheap_insert(c.ng, &ss->ng_count, NodeDist{istage[r], dv}, &ss->oflow);
             }
             /* dynamic pivots (BKTIndex.cpp:346-349) */
             ss->want_tree = (ndheap_top(c.ng, ss->ng_count).distance >
