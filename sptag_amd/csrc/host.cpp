@@ -20,6 +20,7 @@
 #include <hip/hip_runtime.h>
 
 #include <cstdio>
+#include <cstdlib>
 #include <cstring>
 #include <mutex>
 #include <string>
@@ -344,7 +345,7 @@ static int search_device_core(SptagAmdIndex* ix, const void* d_q, int32_t nq,
      * several workgroups fit per CU; the rare query that outgrows them is
      * rerun on the global-heap variant at the reference's own capacities
      * (WorkSpace.h:265) — a speed tradeoff, never a semantic one. */
-    cfg.ng_cap = max_check / 2;
+    cfg.ng_cap = max_check / 2 + 512;
     cfg.spt_cap = 4096;
 
     int lds_limit = 64 * 1024;
@@ -407,6 +408,9 @@ static int search_device_core(SptagAmdIndex* ix, const void* d_q, int32_t nq,
                 ix->last_popped += stats[2 * (size_t)i + 1];
             }
         }
+        if (!redo.empty() && getenv("SPTAG_AMD_DEBUG"))
+            fprintf(stderr, "sptag_amd: %zu/%d queries overflowed (rerun)\n",
+                    redo.size(), nq);
         if (redo.size() * 2 > (size_t)nq) {
             all_global = true;   /* cheaper to redo the whole batch */
             redo.clear();

@@ -311,24 +311,29 @@ DEV float ref_dist_grp_i8(const int8_t* __restrict__ q,
 }
 
 /* load/compute split of the fixed-dim distance, for software-pipelining
- * candidate rounds (loads of round r+1 issue while round r reduces). */
+ * candidate rounds (loads of round r+1 issue while round r reduces).
+ * Only the candidate's elements are register-resident: the query side is
+ * re-read from LDS in the reduce (cheap ds_reads), keeping the pipeline's
+ * register cost at C floats per fragment. */
 template <int DM, int DFIX>
 struct FragF32 {
     static constexpr int C = DFIX / 16;
-    float x[C], y[C];
-    DEV void load(const float* __restrict__ q, const float* __restrict__ v)
+    float y[C];
+    DEV void load(const float* __restrict__ v)
     {
         const int g = threadIdx.x & 15;
 #pragma unroll
-        for (int c = 0; c < C; c++) { x[c] = q[c * 16 + g]; y[c] = v[c * 16 + g]; }
+        for (int c = 0; c < C; c++) y[c] = v[c * 16 + g];
     }
-    DEV float reduce() const
+    DEV float reduce(const float* __restrict__ q) const
     {
+        const int g = threadIdx.x & 15;
         float a = 0.0f;
 #pragma unroll
-        for (int c = 0; c < C; c++)
-            a = (DM == DM_L2) ? fmaf(x[c] - y[c], x[c] - y[c], a)
-                              : fmaf(x[c], y[c], a);
+        for (int c = 0; c < C; c++) {
+            float x = q[c * 16 + g];
+            a = (DM == DM_L2) ? fmaf(x - y[c], x - y[c], a) : fmaf(x, y[c], a);
+        }
         a = a + __shfl_down(a, 8, 16);
         a = a + __shfl_down(a, 4, 16);
         float sv = ((__shfl(a, 0, 16) + __shfl(a, 1, 16)) + __shfl(a, 2, 16)) +
@@ -398,11 +403,11 @@ DEV void stage_dists_pipe_f32(QCtx<float>& c, int cnt)
     const int grp = c.lane >> 4;
     FragF32<DM, DFIX> A, B;
     int j = grp;
-    if (j < cnt) A.load(c.qlds, vec_at<float>(di, c.istage[j]));
+    if (j < cnt) A.load(vec_at<float>(di, c.istage[j]));
     while (j < cnt) {
         int jn = j + 4;
-        if (jn < cnt) B.load(c.qlds, vec_at<float>(di, c.istage[jn]));
-        float dv = A.reduce();
+        if (jn < cnt) B.load(vec_at<float>(di, c.istage[jn]));
+        float dv = A.reduce(c.qlds);
         if ((c.lane & 15) == 0) c.dstage[j] = dv;
         A = B;
         j = jn;
@@ -418,12 +423,12 @@ DEV void stage_dists(QCtx<T>& c, int cnt)
     if constexpr (sizeof(T) == 4) {
         QCtx<float>& cf = reinterpret_cast<QCtx<float>&>(c);
         switch (c.di->dim) {
+        /* pipelining holds 2 fragments (2*C VGPRs) live; C <= 8 only */
         case 32:  stage_dists_pipe_f32<DM, 32>(cf, cnt); return;
         case 48:  stage_dists_pipe_f32<DM, 48>(cf, cnt); return;
         case 64:  stage_dists_pipe_f32<DM, 64>(cf, cnt); return;
         case 96:  stage_dists_pipe_f32<DM, 96>(cf, cnt); return;
         case 128: stage_dists_pipe_f32<DM, 128>(cf, cnt); return;
-        case 256: stage_dists_pipe_f32<DM, 256>(cf, cnt); return;
         default: break;
         }
     }
