@@ -18,7 +18,9 @@ import numpy as np
 __all__ = ["AnnIndex", "load_library", "gpu_available", "SptagAmdError"]
 
 _LIB = None
-_LIBPATH = os.path.join(os.path.dirname(os.path.abspath(__file__)), "libsptag_amd.so")
+_LIBPATH = os.environ.get(
+    "SPTAG_AMD_LIB",
+    os.path.join(os.path.dirname(os.path.abspath(__file__)), "libsptag_amd.so"))
 
 VT_FLOAT, VT_INT8 = 0, 1
 DM_L2, DM_COSINE = 0, 1

@@ -346,6 +346,8 @@ static int search_device_core(SptagAmdIndex* ix, const void* d_q, int32_t nq,
      * rerun on the global-heap variant at the reference's own capacities
      * (WorkSpace.h:265) — a speed tradeoff, never a semantic one. */
     cfg.ng_cap = max_check / 2 + 512;
+    if (const char* e = getenv("SPTAG_AMD_NG_CAP"))   /* perf experiments */
+        cfg.ng_cap = std::max(256, atoi(e));
     cfg.spt_cap = 4096;
 
     int lds_limit = 64 * 1024;

@@ -525,8 +525,12 @@ DEV void init_search_trees_dev(QCtx<T>& c)
  * main search kernel
  * ------------------------------------------------------------------ */
 
+#ifndef SPTAG_LB_WAVES
+#define SPTAG_LB_WAVES 1   /* min waves/SIMD hint; raise to cap VGPRs */
+#endif
+
 template <typename T, int DM, bool LDSHEAP>
-__global__ __launch_bounds__(64)
+__global__ __launch_bounds__(64, SPTAG_LB_WAVES)
 void bkt_search_kernel(DevIndex di, SearchCfg cfg, SearchBufs bufs)
 {
     const int q = blockIdx.x;
