@@ -51,6 +51,13 @@ CONFIGS = {
     "bkt_100m_d100_i8_cos": dict(n=100_000_000, d=100, dtype="i8",
                                  metric="Cosine", nq=10_000, k=10,
                                  ncenters=16384, sigma=30.0),
+    # int8 validation scale (same dtype/metric path as config #3)
+    "bkt_10m_d100_i8_cos": dict(n=10_000_000, d=100, dtype="i8",
+                                metric="Cosine", nq=10_000, k=10,
+                                ncenters=8192, sigma=30.0),
+    "bkt_10m_d100_i8_l2": dict(n=10_000_000, d=100, dtype="i8",
+                               metric="L2", nq=10_000, k=10,
+                               ncenters=8192, sigma=30.0),
 }
 DEFAULT_WORKLOAD = "bkt_10m_d128_f32_l2"
 MC_SWEEP = [512, 1024, 2048, 4096, 8192, 16384]
