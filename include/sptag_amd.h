@@ -77,6 +77,15 @@ SptagAmdIndex* sptag_amd_create_index(int32_t n, int32_t dim, int valuetype,
                                       const uint8_t* deleted /* NULL = none */,
                                       int device);
 
+/* KDT variant of create (reference inc/Core/Common/KDTree.h:22 KDTNode
+ * {left,right,split_dim,split_value}; kdt_nodes = 16-byte records). */
+SptagAmdIndex* sptag_amd_create_index_kdt(int32_t n, int32_t dim, int valuetype,
+                                          int distmethod, const void* vectors,
+                                          int32_t ntrees, const int32_t* tree_start,
+                                          int32_t n_tree_nodes, const void* kdt_nodes,
+                                          int32_t degree, const int32_t* graph,
+                                          const uint8_t* deleted, int device);
+
 void sptag_amd_free_index(SptagAmdIndex* idx);
 
 /* Batched search. queries: nq contiguous vectors of the index's dtype/dim.

@@ -50,7 +50,8 @@ def run(cmd):
     subprocess.run(cmd, check=True, cwd=REPO)
 
 
-def gen_fixture(name, data, queries, vtype, dist, maxchecks=MAXCHECKS, k=K):
+def gen_fixture(name, data, queries, vtype, dist, maxchecks=MAXCHECKS, k=K,
+                algo="BKT"):
     d = os.path.join(GOLDEN, name)
     idx = os.path.join(d, "index")
     os.makedirs(d, exist_ok=True)
@@ -59,7 +60,7 @@ def gen_fixture(name, data, queries, vtype, dist, maxchecks=MAXCHECKS, k=K):
 
     run([os.path.join(REF, "indexbuilder"),
          "-d", str(data.shape[1]), "-v", vtype, "-f", "DEFAULT",
-         "-i", os.path.join(d, "data.bin"), "-o", idx, "-a", "BKT",
+         "-i", os.path.join(d, "data.bin"), "-o", idx, "-a", algo,
          "-t", "4",
          "Index.DistCalcMethod=" + dist])
 
@@ -75,7 +76,7 @@ def gen_fixture(name, data, queries, vtype, dist, maxchecks=MAXCHECKS, k=K):
     os.remove(os.path.join(d, "data.bin"))
     with open(os.path.join(d, "meta.json"), "w") as f:
         json.dump({"n": int(data.shape[0]), "dim": int(data.shape[1]),
-                   "valuetype": vtype, "distmethod": dist, "k": k,
+                   "valuetype": vtype, "distmethod": dist, "k": k, "algo": algo,
                    "maxchecks": maxchecks, "nq": int(queries.shape[0])}, f)
     print(f"fixture {name} done")
 
@@ -120,6 +121,21 @@ def main():
     queries = rng.random((100, 16), dtype=np.float32)
     gen_fixture("f32_l2_dups", data, queries, "Float", "L2",
                 maxchecks=[512, 2048], k=10)
+
+    # 7. KDT fixtures (config #4 algo): same traversal machinery, kd-tree
+    #    seeds + no-better-propagation termination (KDTIndex.cpp:184-241)
+    data = rng.standard_normal((10000, 64)).astype(np.float32)
+    queries = rng.standard_normal((100, 64)).astype(np.float32)
+    gen_fixture("kdt_f32_cos_n10k_d64", data, queries, "Float", "Cosine",
+                maxchecks=[512, 2048, 8192], algo="KDT")
+    data = rng.random((10000, 32), dtype=np.float32)
+    queries = rng.random((100, 32), dtype=np.float32)
+    gen_fixture("kdt_f32_l2_n10k_d32", data, queries, "Float", "L2",
+                maxchecks=[512, 2048, 8192], algo="KDT")
+    data = rng.integers(-100, 101, (10000, 100)).astype(np.int8)
+    queries = rng.integers(-100, 101, (100, 100)).astype(np.int8)
+    gen_fixture("kdt_i8_l2_n10k_d100", data, queries, "Int8", "L2",
+                maxchecks=[2048, 8192], algo="KDT")
 
     # 6. f32 cosine (config #4 metric; KDT later — BKT for now), normalized
     #    gaussian rows, d=48 not a multiple of 16 to cover distance tails.

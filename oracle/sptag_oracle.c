@@ -115,8 +115,10 @@ float orc_distance(int vt, int dm, const void* x, const void* y, int32_t dim)
  * ------------------------------------------------------------------ */
 
 typedef struct { int32_t centerid, childStart, childEnd; } BktNode; /* BKTree.h:25 */
+typedef struct { int32_t left, right, split_dim; float split_value; } KdtNode; /* KDTree.h:22 */
 
 struct OrcIndex {
+    int algo;                  /* 0 = BKT, 1 = KDT */
     int32_t n, dim;
     int valuetype, distmethod;
     size_t esz;                /* element size in bytes */
@@ -124,7 +126,8 @@ struct OrcIndex {
     int32_t ntrees;
     int32_t* tree_start;
     int32_t n_tree_nodes;
-    BktNode* tree;
+    BktNode* tree;             /* BKT */
+    KdtNode* kdt;              /* KDT */
     int32_t deg;
     int32_t* graph;            /* n*deg row-major */
     uint8_t* deleted;          /* may be NULL -> none */
@@ -386,7 +389,8 @@ typedef struct {
     NDHeap spt;     /* m_SPTQueue, capacity 10*maxCheck */
     DistPQ results; /* m_Results,  capacity max(maxCheck/16, k) */
     VisitSet visited;
-    int checked;    /* m_iNumberOfCheckedLeaves */
+    int checked;       /* m_iNumberOfCheckedLeaves */
+    int tree_checked;  /* m_iNumberOfTreeCheckedLeaves (KDT) */
     int max_check;
 } SearchSpace;
 
@@ -445,9 +449,14 @@ static void search_trees(const OrcIndex* ix, const void* q, SearchSpace* sp, int
 /* default parameters (BKT/ParameterDefinitionList.h:47-49) */
 enum { ORC_INIT_PIVOTS = 50, ORC_OTHER_PIVOTS = 4 };
 
+static int32_t orc_search_kdt(const OrcIndex* ix, const void* q, int32_t k,
+                              int32_t max_check, int32_t* out_vids, float* out_dists);
+
 int32_t orc_search(const OrcIndex* ix, const void* q, int32_t k,
                    int32_t max_check, int32_t* out_vids, float* out_dists)
 {
+    if (ix->algo == 1)
+        return orc_search_kdt(ix, q, k, max_check, out_vids, out_dists);
     SearchSpace sp;
     ndheap_init(&sp.ng, max_check * 30);
     ndheap_init(&sp.spt, max_check * 10);
@@ -455,6 +464,7 @@ int32_t orc_search(const OrcIndex* ix, const void* q, int32_t k,
     distpq_init(&sp.results, res_cap);
     vset_init(&sp.visited, max_check * 2 + 64 * 32);
     sp.checked = 0;
+    sp.tree_checked = 0;
     sp.max_check = max_check;
 
     QRes* storage = (QRes*)malloc(sizeof(QRes) * (size_t)k);
@@ -518,6 +528,119 @@ int32_t orc_search(const OrcIndex* ix, const void* q, int32_t k,
     qrs_sort(&query);
     for (int i = 0; i < k; i++) { out_vids[i] = query.r[i].vid; out_dists[i] = query.r[i].dist; }
 
+    int32_t checked = sp.checked;
+    free(storage);
+    vset_destroy(&sp.visited);
+    distpq_destroy(&sp.results);
+    ndheap_destroy(&sp.spt);
+    ndheap_destroy(&sp.ng);
+    return checked;
+}
+
+/* ------------------------------------------------------------------ *
+ * KDT search — exact restatement of KDT::Index<T>::Search<Q,...>
+ * (src/Core/KDT/KDTIndex.cpp:184-241) and KDTree::InitSearchTrees /
+ * SearchTrees / KDTSearch (inc/Core/Common/KDTree.h:213-273).
+ * ------------------------------------------------------------------ */
+
+enum { ORC_KDT_NOBETTER_THRESHOLD = 3 };  /* KDT/ParameterDefinitionList.h */
+
+/* KDTSearch (KDTree.h:234-271): iterative form of the tail recursion.
+ * distBound accumulates squared split-plane distances (lower bound);
+ * the off-path child goes to the SPT queue with the tightened bound. */
+static void kdt_search_node(const OrcIndex* ix, const void* q, SearchSpace* sp,
+                            int32_t node, float dist_bound)
+{
+    while (node >= 0) {
+        const KdtNode* tn = &ix->kdt[node];
+        float qv;
+        if (ix->valuetype == ORC_VT_FLOAT)
+            qv = ((const float*)q)[tn->split_dim];
+        else
+            qv = (float)((const int8_t*)q)[tn->split_dim];
+        float diff = qv - tn->split_value;
+        float other_bound = dist_bound + diff * diff;
+        int32_t best = diff < 0 ? tn->left : tn->right;
+        int32_t other = diff < 0 ? tn->right : tn->left;
+        NodeDist nd = { other, other_bound };
+        ndheap_insert(&sp->spt, nd);
+        node = best;
+    }
+    /* leaf: node < 0 encodes -(vector id + 1) */
+    int32_t index = -node - 1;
+    if (index >= ix->n) return;
+    if (vset_check_and_set(&sp->visited, index)) return;
+    sp->tree_checked++;
+    sp->checked++;
+    NodeDist nd = { index, idx_dist(ix, q, index) };
+    ndheap_insert(&sp->ng, nd);
+}
+
+static void kdt_search_trees(const OrcIndex* ix, const void* q, SearchSpace* sp,
+                             int limits)
+{
+    while (sp->spt.count > 0 && sp->checked < limits) {
+        NodeDist tcell = ndheap_pop(&sp->spt);
+        kdt_search_node(ix, q, sp, tcell.node, tcell.distance);
+    }
+}
+
+static int32_t orc_search_kdt(const OrcIndex* ix, const void* q, int32_t k,
+                              int32_t max_check, int32_t* out_vids, float* out_dists)
+{
+    SearchSpace sp;
+    ndheap_init(&sp.ng, max_check * 30);
+    ndheap_init(&sp.spt, max_check * 10);
+    distpq_init(&sp.results, max_check / 16 > k ? max_check / 16 : k);
+    vset_init(&sp.visited, max_check * 2 + 64 * 32);
+    sp.checked = 0;
+    sp.tree_checked = 0;
+    sp.max_check = max_check;
+
+    QRes* storage = (QRes*)malloc(sizeof(QRes) * (size_t)k);
+    QResultSet query;
+    qrs_init(&query, storage, k);
+
+    for (int t = 0; t < ix->ntrees; t++)
+        kdt_search_node(ix, q, &sp, ix->tree_start[t], 0.0f);
+    kdt_search_trees(ix, q, &sp, ORC_INIT_PIVOTS);
+
+    int no_better = 0;   /* m_iNumOfContinuousNoBetterPropagation */
+    while (sp.ng.count > 0) {
+        NodeDist gnode = ndheap_pop(&sp.ng);
+        const int32_t* row = ix->graph + (size_t)gnode.node * ix->deg;
+        if (not_deleted(ix, gnode.node)) {
+            if (!qrs_add(&query, gnode.node, gnode.distance) &&
+                sp.checked > sp.max_check) {
+                break;
+            }
+        }
+        float worst = qrs_worst(&query);
+        float upper_bound = worst > gnode.distance ? worst : gnode.distance;
+        int local_opt = 1;
+        for (int i = 0; i < ix->deg; i++) {
+            int32_t nn = row[i];
+            if (nn < 0) break;
+            if (vset_check_and_set(&sp.visited, nn)) continue;
+            float d = idx_dist(ix, q, nn);
+            if (d <= upper_bound) local_opt = 0;
+            sp.checked++;
+            NodeDist nd = { nn, d };
+            ndheap_insert(&sp.ng, nd);
+        }
+        if (local_opt) no_better++;
+        else no_better = 0;
+        if (no_better > ORC_KDT_NOBETTER_THRESHOLD) {
+            if (sp.tree_checked <= sp.checked / 10) {
+                kdt_search_trees(ix, q, &sp, ORC_OTHER_PIVOTS + sp.checked);
+            } else if (gnode.distance > qrs_worst(&query)) {
+                break;
+            }
+        }
+    }
+
+    qrs_sort(&query);
+    for (int i = 0; i < k; i++) { out_vids[i] = query.r[i].vid; out_dists[i] = query.r[i].dist; }
     int32_t checked = sp.checked;
     free(storage);
     vset_destroy(&sp.visited);
@@ -610,10 +733,42 @@ OrcIndex* orc_create_index(int32_t n, int32_t dim, int valuetype, int distmethod
     return ix;
 }
 
+OrcIndex* orc_create_kdt_index(int32_t n, int32_t dim, int valuetype, int distmethod,
+                               const void* vectors,
+                               int32_t ntrees, const int32_t* tree_start,
+                               int32_t n_tree_nodes, const void* kdt_nodes,
+                               int32_t degree, const int32_t* graph,
+                               const uint8_t* deleted)
+{
+    OrcIndex* ix = (OrcIndex*)calloc(1, sizeof(OrcIndex));
+    ix->algo = 1;
+    ix->n = n; ix->dim = dim;
+    ix->valuetype = valuetype; ix->distmethod = distmethod;
+    ix->esz = valuetype == ORC_VT_FLOAT ? 4 : 1;
+    size_t vbytes = (size_t)n * dim * ix->esz;
+    ix->vectors = malloc(vbytes);
+    memcpy(ix->vectors, vectors, vbytes);
+    ix->ntrees = ntrees;
+    ix->tree_start = (int32_t*)malloc(sizeof(int32_t) * (size_t)ntrees);
+    memcpy(ix->tree_start, tree_start, sizeof(int32_t) * (size_t)ntrees);
+    ix->n_tree_nodes = n_tree_nodes;
+    ix->kdt = (KdtNode*)malloc(sizeof(KdtNode) * (size_t)n_tree_nodes);
+    memcpy(ix->kdt, kdt_nodes, sizeof(KdtNode) * (size_t)n_tree_nodes);
+    ix->deg = degree;
+    ix->graph = (int32_t*)malloc(sizeof(int32_t) * (size_t)n * degree);
+    memcpy(ix->graph, graph, sizeof(int32_t) * (size_t)n * degree);
+    if (deleted) {
+        ix->deleted = (uint8_t*)malloc((size_t)n);
+        memcpy(ix->deleted, deleted, (size_t)n);
+        for (int32_t i = 0; i < n; i++) ix->deleted_count += deleted[i] ? 1 : 0;
+    }
+    return ix;
+}
+
 void orc_free_index(OrcIndex* ix)
 {
     if (!ix) return;
-    free(ix->vectors); free(ix->tree_start); free(ix->tree);
+    free(ix->vectors); free(ix->tree_start); free(ix->tree); free(ix->kdt);
     free(ix->graph); free(ix->deleted); free(ix);
 }
 
@@ -680,11 +835,15 @@ OrcIndex* orc_load_index(const char* folder)
     ini = (char*)realloc(ini, tsz + 1);
     ini[tsz] = 0;
 
-    int vt = ORC_VT_FLOAT, dm = ORC_DM_L2;
-    if (ini_get(ini, "Index", "IndexAlgoType", val, sizeof val) && strcmp(val, "BKT") != 0) {
-        fprintf(stderr, "oracle: only BKT supported, got %s\n", val);
-        free(ini);
-        return NULL;
+    int vt = ORC_VT_FLOAT, dm = ORC_DM_L2, algo = 0;
+    if (ini_get(ini, "Index", "IndexAlgoType", val, sizeof val)) {
+        if (strcmp(val, "BKT") == 0) algo = 0;
+        else if (strcmp(val, "KDT") == 0) algo = 1;
+        else {
+            fprintf(stderr, "oracle: unsupported IndexAlgoType %s\n", val);
+            free(ini);
+            return NULL;
+        }
     }
     if (ini_get(ini, "Index", "ValueType", val, sizeof val)) {
         if (strcmp(val, "Float") == 0) vt = ORC_VT_FLOAT;
@@ -736,9 +895,18 @@ OrcIndex* orc_load_index(const char* folder)
     if (vt == ORC_VT_INT8 && (int64_t)dim * 254 * 254 >= (1ll << 24))
         fprintf(stderr, "oracle: warning: int8 dim %d may lose float exactness\n", dim);
 
-    OrcIndex* ix = orc_create_index(n, dim, vt, dm, vb + 8,
-                                    ntrees, tstart, nnodes, tnodes,
-                                    deg, (int32_t*)(gb + 8), del);
+    OrcIndex* ix;
+    if (algo == 0)
+        ix = orc_create_index(n, dim, vt, dm, vb + 8, ntrees, tstart, nnodes,
+                              tnodes, deg, (int32_t*)(gb + 8), del);
+    else {
+        /* KDT tree.bin: [int32 #trees][int32 starts x #][int32 count]
+         * [KDTNode{left,right,split_dim,split_value} x count]
+         * (KDTree.h:123-135 SaveTrees / :190-201 LoadTrees) */
+        const KdtNode* knodes = (const KdtNode*)(tb + 8 + 4 * (size_t)ntrees);
+        ix = orc_create_kdt_index(n, dim, vt, dm, vb + 8, ntrees, tstart,
+                                  nnodes, knodes, deg, (int32_t*)(gb + 8), del);
+    }
     ix->deleted_count = delcount;
     free(vb); free(tb); free(gb); free(db);
     return ix;

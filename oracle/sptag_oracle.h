@@ -44,6 +44,15 @@ OrcIndex* orc_create_index(int32_t n, int32_t dim, int valuetype, int distmethod
                            int32_t degree, const int32_t* graph,
                            const uint8_t* deleted /* may be NULL */);
 
+/* KDT variant (reference inc/Core/Common/KDTree.h:22 KDTNode
+ * {left,right,split_dim,split_value}); kdt_nodes is 16-byte records. */
+OrcIndex* orc_create_kdt_index(int32_t n, int32_t dim, int valuetype, int distmethod,
+                               const void* vectors,
+                               int32_t ntrees, const int32_t* tree_start,
+                               int32_t n_tree_nodes, const void* kdt_nodes,
+                               int32_t degree, const int32_t* graph,
+                               const uint8_t* deleted);
+
 void orc_free_index(OrcIndex* idx);
 
 int32_t orc_num_vectors(const OrcIndex* idx);

@@ -56,6 +56,8 @@ def load_library():
         ctypes.c_void_p, ctypes.c_int32, ctypes.c_void_p, ctypes.c_int32,
         ctypes.c_void_p, ctypes.c_int32, ctypes.c_void_p, ctypes.c_void_p,
         ctypes.c_int]
+    lib.sptag_amd_create_index_kdt.restype = ctypes.c_void_p
+    lib.sptag_amd_create_index_kdt.argtypes = lib.sptag_amd_create_index.argtypes
     lib.sptag_amd_free_index.argtypes = [ctypes.c_void_p]
     lib.sptag_amd_search_batch.restype = ctypes.c_int
     lib.sptag_amd_search_batch.argtypes = [
@@ -136,6 +138,31 @@ class AnnIndex:
             n, dim, vt, dm, vectors.ctypes.data_as(ctypes.c_void_p),
             len(tree_start), tree_start.ctypes.data_as(ctypes.c_void_p),
             tree_nodes.size // 3, tree_nodes.ctypes.data_as(ctypes.c_void_p),
+            graph.shape[1], graph.ctypes.data_as(ctypes.c_void_p),
+            delp, device)
+        return cls(h)
+
+    @classmethod
+    def FromArraysKDT(cls, vectors, tree_start, kdt_nodes, graph, distmethod,
+                      deleted=None, device=0):
+        """KDT variant: kdt_nodes is an int32 [N,4] array whose 4th column
+        holds the float split_value bit pattern (KDTree.h:22)."""
+        lib = load_library()
+        vectors = np.ascontiguousarray(vectors)
+        vt = VT_FLOAT if vectors.dtype == np.float32 else VT_INT8
+        n, dim = vectors.shape
+        tree_start = np.ascontiguousarray(tree_start, dtype=np.int32)
+        kdt_nodes = np.ascontiguousarray(kdt_nodes, dtype=np.int32)
+        graph = np.ascontiguousarray(graph, dtype=np.int32)
+        dm = {"L2": DM_L2, "Cosine": DM_COSINE}.get(distmethod, distmethod)
+        delp = None
+        if deleted is not None:
+            deleted = np.ascontiguousarray(deleted, dtype=np.uint8)
+            delp = deleted.ctypes.data_as(ctypes.c_void_p)
+        h = lib.sptag_amd_create_index_kdt(
+            n, dim, vt, dm, vectors.ctypes.data_as(ctypes.c_void_p),
+            len(tree_start), tree_start.ctypes.data_as(ctypes.c_void_p),
+            kdt_nodes.size // 4, kdt_nodes.ctypes.data_as(ctypes.c_void_p),
             graph.shape[1], graph.ctypes.data_as(ctypes.c_void_p),
             delp, device)
         return cls(h)

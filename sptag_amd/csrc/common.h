@@ -15,21 +15,26 @@ enum { DEFAULT_MAXCHECK = 8192, INIT_PIVOTS = 50, OTHER_PIVOTS = 4 };
 /* hard caps of the v1 kernel (host validates before launch) */
 enum { MAX_DEG = 64, MAX_K = 64, MAX_DIM = 4096 };
 
+enum { ALGO_BKT = 0, ALGO_KDT = 1 };
+
 /* device-resident index description (POD, passed by value to kernels) */
 struct DevIndex {
     const void* vectors;       /* n*dim row-major, element size by vt */
     const int32_t* graph;      /* n*deg row-major adjacency */
-    const int32_t* tree_nodes; /* 3*int32 {centerid,childStart,childEnd} per node */
+    const int32_t* tree_nodes; /* BKT: {centerid,childStart,childEnd} int32x3;
+                                  KDT: {left,right,split_dim,split_value} 16B */
     const int32_t* tree_start; /* ntrees roots */
     const uint8_t* deleted;    /* may be null */
     int32_t n, dim, deg, ntrees, n_tree_nodes;
     int32_t has_deleted;
+    int32_t algo;
 };
 
 /* per-launch search configuration */
 struct SearchCfg {
     int32_t nq, k, max_check;
     int32_t init_pivots, other_pivots;
+    int32_t nobetter_threshold;  /* KDT: ThresholdOfNumberOfContinuousNoBetterPropagation */
     int32_t ng_cap, spt_cap, dpq_cap;  /* heap capacities (entries) */
     int32_t vcap;                      /* visited table slots (pow2) */
 };

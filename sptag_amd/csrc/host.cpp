@@ -34,10 +34,12 @@ namespace {
 
 struct HostIndex {
     int device = 0;
+    int algo = ALGO_BKT;
     int vt = VT_FLOAT, dm = DM_L2;
     int32_t n = 0, dim = 0, deg = 0, ntrees = 0, n_tree_nodes = 0;
     int32_t default_maxcheck = DEFAULT_MAXCHECK;
     int32_t init_pivots = INIT_PIVOTS, other_pivots = OTHER_PIVOTS;
+    int32_t nobetter_threshold = 3;  /* KDT ParameterDefinitionList */
     bool has_deleted = false;
     /* device allocations */
     void* d_vectors = nullptr;
@@ -74,6 +76,7 @@ struct HostIndex {
         di.n = n; di.dim = dim; di.deg = deg; di.ntrees = ntrees;
         di.n_tree_nodes = n_tree_nodes;
         di.has_deleted = has_deleted ? 1 : 0;
+        di.algo = algo;
         return di;
     }
 };
@@ -221,6 +224,44 @@ SptagAmdIndex* sptag_amd_create_index(int32_t n, int32_t dim, int valuetype,
     return ix;
 }
 
+SptagAmdIndex* sptag_amd_create_index_kdt(int32_t n, int32_t dim, int valuetype,
+                                          int distmethod, const void* vectors,
+                                          int32_t ntrees, const int32_t* tree_start,
+                                          int32_t n_tree_nodes, const void* kdt_nodes,
+                                          int32_t degree, const int32_t* graph,
+                                          const uint8_t* deleted, int device)
+{
+    if (n <= 0 || dim <= 0 || dim > MAX_DIM || degree <= 0 || degree > MAX_DEG ||
+        ntrees <= 0 || n_tree_nodes <= 0)
+        return nullptr;
+    auto* ix = new SptagAmdIndex();
+    ix->device = device;
+    ix->algo = ALGO_KDT;
+    ix->vt = valuetype;
+    ix->dm = distmethod;
+    ix->n = n; ix->dim = dim; ix->deg = degree;
+    ix->ntrees = ntrees;
+    ix->n_tree_nodes = n_tree_nodes;
+    ix->h_vectors.assign((const char*)vectors,
+                         (const char*)vectors + (size_t)n * dim * ix->esz());
+    ix->h_graph.assign(graph, graph + (size_t)n * degree);
+    ix->h_tree_start.assign(tree_start, tree_start + ntrees);
+    const int32_t* kw = (const int32_t*)kdt_nodes;
+    ix->h_tree.assign(kw, kw + (size_t)n_tree_nodes * 4);
+    if (deleted) {
+        ix->h_deleted.assign(deleted, deleted + n);
+        for (int32_t i = 0; i < n; i++) ix->deleted_count += deleted[i] ? 1 : 0;
+        ix->has_deleted = ix->deleted_count > 0;
+    }
+    if (sptag_amd_gpu_available()) {
+        if (upload(ix) != SPTAG_AMD_OK) {
+            sptag_amd_free_index(ix);
+            return nullptr;
+        }
+    }
+    return ix;
+}
+
 SptagAmdIndex* sptag_amd_load_index(const char* folder, int device)
 {
     std::string dir(folder);
@@ -231,10 +272,15 @@ SptagAmdIndex* sptag_amd_load_index(const char* folder, int device)
     std::string ini(raw.begin(), raw.end());
 
     std::string val;
-    if (ini_get(ini, "Index", "IndexAlgoType", val) && val != "BKT") {
-        fprintf(stderr, "sptag_amd: IndexAlgoType %s not supported yet (BKT only)\n",
-                val.c_str());
-        return nullptr;
+    int algo = ALGO_BKT;
+    if (ini_get(ini, "Index", "IndexAlgoType", val)) {
+        if (val == "BKT") algo = ALGO_BKT;
+        else if (val == "KDT") algo = ALGO_KDT;
+        else {
+            fprintf(stderr, "sptag_amd: IndexAlgoType %s not supported\n",
+                    val.c_str());
+            return nullptr;
+        }
     }
     int vt = VT_FLOAT, dm = DM_L2;
     if (ini_get(ini, "Index", "ValueType", val)) {
@@ -279,9 +325,17 @@ SptagAmdIndex* sptag_amd_load_index(const char* folder, int device)
         if (delcount > 0) del = (const uint8_t*)db.data() + 12;
     }
 
-    auto* ix = sptag_amd_create_index(n, dim, vt, dm, vb.data() + 8, ntrees, tstart,
-                                      nnodes, tnodes, deg, gh + 2, del, device);
+    SptagAmdIndex* ix;
+    if (algo == ALGO_BKT)
+        ix = sptag_amd_create_index(n, dim, vt, dm, vb.data() + 8, ntrees, tstart,
+                                    nnodes, tnodes, deg, gh + 2, del, device);
+    else
+        ix = sptag_amd_create_index_kdt(n, dim, vt, dm, vb.data() + 8, ntrees,
+                                        tstart, nnodes, tnodes, deg, gh + 2, del,
+                                        device);
     if (!ix) return nullptr;
+    if (ini_get(ini, "Index", "ThresholdOfNumberOfContinuousNoBetterPropagation", val))
+        ix->nobetter_threshold = atoi(val.c_str());
     if (ini_get(ini, "Index", "MaxCheck", val)) ix->default_maxcheck = atoi(val.c_str());
     if (ini_get(ini, "Index", "NumberOfInitialDynamicPivots", val))
         ix->init_pivots = atoi(val.c_str());
@@ -339,13 +393,16 @@ static int search_device_core(SptagAmdIndex* ix, const void* d_q, int32_t nq,
     cfg.max_check = max_check;
     cfg.init_pivots = ix->init_pivots;
     cfg.other_pivots = ix->other_pivots;
+    cfg.nobetter_threshold = ix->nobetter_threshold;
     cfg.dpq_cap = std::max(max_check / 16, k);   /* WorkSpace.h:268 */
     cfg.vcap = (int32_t)next_pow2((uint32_t)std::max(4096, max_check * 4));
     /* LDS-variant capacities: sized to the TYPICAL traversal occupancy so
      * several workgroups fit per CU; the rare query that outgrows them is
      * rerun on the global-heap variant at the reference's own capacities
      * (WorkSpace.h:265) — a speed tradeoff, never a semantic one. */
-    cfg.ng_cap = max_check / 2 + 512;
+    /* KDT inserts every computed neighbor into the frontier (no results-
+     * queue gate), so its occupancy tracks `checked` itself. */
+    cfg.ng_cap = ix->algo == ALGO_KDT ? max_check + 2048 : max_check / 2 + 512;
     if (const char* e = getenv("SPTAG_AMD_NG_CAP"))   /* perf experiments */
         cfg.ng_cap = std::max(256, atoi(e));
     cfg.spt_cap = 4096;
@@ -600,7 +657,8 @@ int sptag_amd_save_index(SptagAmdIndex* ix, const char* folder)
         fwrite(&ix->ntrees, 4, 1, f);
         fwrite(ix->h_tree_start.data(), 4, ix->ntrees, f);
         fwrite(&ix->n_tree_nodes, 4, 1, f);
-        fwrite(ix->h_tree.data(), 12, ix->n_tree_nodes, f);
+        fwrite(ix->h_tree.data(), ix->algo == ALGO_KDT ? 16 : 12,
+               ix->n_tree_nodes, f);
         fclose(f);
     }
     int32_t gh[2] = {ix->n, ix->deg};
@@ -616,6 +674,28 @@ int sptag_amd_save_index(SptagAmdIndex* ix, const char* folder)
     {
         FILE* f = fopen((dir + "indexloader.ini").c_str(), "wb");
         if (!f) return SPTAG_AMD_ERR_IO;
+        if (ix->algo == ALGO_KDT) {
+            fprintf(f,
+                    "[Index]\nIndexAlgoType=KDT\nValueType=%s\n\n"
+                    "TreeFilePath=tree.bin\nGraphFilePath=graph.bin\n"
+                    "VectorFilePath=vectors.bin\nDeleteVectorFilePath=deletes.bin\n"
+                    "KDTNumber=%d\nNumTopDimensionKDTSplit=5\nSamples=100\n"
+                    "TPTNumber=32\nTPTLeafSize=2000\nNumTopDimensionTPTSplit=5\n"
+                    "NeighborhoodSize=%d\nGraphNeighborhoodScale=2.000000\n"
+                    "GraphCEFScale=2.000000\nRefineIterations=2\nCEF=1000\n"
+                    "MaxCheckForRefineGraph=8192\nNumberOfThreads=4\n"
+                    "DistCalcMethod=%s\nMaxCheck=%d\n"
+                    "ThresholdOfNumberOfContinuousNoBetterPropagation=%d\n"
+                    "NumberOfInitialDynamicPivots=%d\n"
+                    "NumberOfOtherDynamicPivots=%d\nHashTableExponent=2\n"
+                    "DataBlockSize=1048576\nDataCapacity=2147483647\n"
+                    "MetaRecordSize=10\n",
+                    ix->vt == VT_FLOAT ? "Float" : "Int8", ix->ntrees, ix->deg,
+                    ix->dm == DM_L2 ? "L2" : "Cosine", ix->default_maxcheck,
+                    ix->nobetter_threshold, ix->init_pivots, ix->other_pivots);
+            fclose(f);
+            return SPTAG_AMD_OK;
+        }
         fprintf(f,
                 "[Index]\n"
                 "IndexAlgoType=BKT\n"

@@ -360,6 +360,8 @@ template <> DEV float ref_dist_grp<int8_t, DM_COSINE>(const int8_t* q, const int
 struct SerialState {          /* in LDS; lane 0 writes, wave reads after sync */
     int ng_count, spt_count, dpq_len, checked;
     int oflow, terminate, break_flag, want_tree;
+    int tree_checked, no_better;   /* KDT counters */
+    float fbcast;                  /* KDT upper-bound broadcast */
     NodeDist popped;
 };
 
@@ -686,6 +688,208 @@ void bkt_search_kernel(DevIndex di, SearchCfg cfg, SearchBufs bufs)
 }
 
 /* ------------------------------------------------------------------ *
+ * KDT search (reference src/Core/KDT/KDTIndex.cpp:184-241 +
+ * inc/Core/Common/KDTree.h:213-273). Shares the heaps/visited/top-k/
+ * distance machinery; differs in the seed descent (kd-tree with
+ * squared-split-plane lower bounds in the SPT queue), in the frontier
+ * update (every computed neighbor enters the queue — no results-queue
+ * gate) and in termination (continuous-no-better-propagation counter).
+ * ------------------------------------------------------------------ */
+
+struct KdtNode { int32_t left, right, split_dim; float split_value; };
+
+/* KDTree.h:234-271 KDTSearch, iterative; lane 0 descends (serial pointer
+ * chase), the wave's first 16-lane group computes the leaf distance. */
+template <typename T, int DM>
+DEV void kdt_search_node_dev(QCtx<T>& c, int32_t node, float dist_bound)
+{
+    const DevIndex& di = *c.di;
+    const KdtNode* kn = (const KdtNode*)di.tree_nodes;
+    SerialState* ss = c.ss;
+    /* descent: lane 0 walks, broadcasting the final leaf via ss->popped */
+    if (c.lane == 0) {
+        while (node >= 0) {
+            KdtNode tn = kn[node];
+            float qv = (float)c.qlds[tn.split_dim];
+            float diff = qv - tn.split_value;
+            float other_bound = dist_bound + diff * diff;
+            int32_t best = diff < 0.0f ? tn.left : tn.right;
+            int32_t other = diff < 0.0f ? tn.right : tn.left;
+            ndheap_insert(c.spt, &ss->spt_count, NodeDist{other, other_bound},
+                          &ss->oflow);
+            node = best;
+        }
+        int32_t index = -node - 1;
+        if (index >= di.n) index = -1;
+        else if (visited_test_insert(c.vtab, c.vmask, index, &ss->oflow)) index = -1;
+        ss->popped.node = index;
+    }
+    __syncthreads();
+    int32_t index = ss->popped.node;
+    if (index >= 0) {
+        float dv = 0.0f;
+        if (c.lane < 16)
+            dv = ref_dist_grp<T, DM>(c.qlds, vec_at<T>(di, index), di.dim);
+        if (c.lane == 0) {
+            ss->tree_checked++;
+            ss->checked++;
+            ndheap_insert(c.ng, &ss->ng_count, NodeDist{index, dv}, &ss->oflow);
+        }
+    }
+    __syncthreads();
+}
+
+/* KDTree.h:219-231 SearchTrees: pop bounds until the leaf budget. */
+template <typename T, int DM>
+DEV void kdt_search_trees_dev(QCtx<T>& c, int limits)
+{
+    SerialState* ss = c.ss;
+    for (;;) {
+        __syncthreads();
+        if (ss->spt_count <= 0 || ss->checked >= limits || ss->oflow) break;
+        if (c.lane == 0) ss->popped = ndheap_pop(c.spt, &ss->spt_count);
+        __syncthreads();
+        NodeDist tcell = ss->popped;
+        kdt_search_node_dev<T, DM>(c, tcell.node, tcell.distance);
+    }
+    __syncthreads();
+}
+
+template <typename T, int DM, bool LDSHEAP>
+__global__ __launch_bounds__(64, SPTAG_LB_WAVES)
+void kdt_search_kernel(DevIndex di, SearchCfg cfg, SearchBufs bufs)
+{
+    const int q = blockIdx.x;
+    if (q >= cfg.nq) return;
+    const int lane = threadIdx.x;
+
+    extern __shared__ char smem[];
+    size_t off = 0;
+    T* qlds = (T*)(smem + off);
+    off += ((size_t)di.dim * sizeof(T) + 15) & ~15ul;
+    float* dstage = (float*)(smem + off); off += MAX_DEG * 4;
+    int32_t* istage = (int32_t*)(smem + off); off += MAX_DEG * 4;
+    QRes* qrs = (QRes*)(smem + off); off += (size_t)cfg.k * 8;
+    float* dpq = (float*)(smem + off); off += ((size_t)cfg.dpq_cap + 1) * 4;
+    SerialState* ss = (SerialState*)(smem + off); off += 64;
+
+    HeapRef ng, spt;
+    ng.cap = cfg.ng_cap;  ng.ref_cap = cfg.max_check * 30;
+    spt.cap = cfg.spt_cap; spt.ref_cap = cfg.max_check * 10;
+    spt.a = (NodeDist*)bufs.gheap_spt + (size_t)q * (cfg.spt_cap + 1);
+    if (LDSHEAP) {
+        ng.a = (NodeDist*)(smem + off); off += ((size_t)cfg.ng_cap + 1) * 8;
+    } else {
+        ng.a = (NodeDist*)bufs.gheap_ng + (size_t)q * (cfg.ng_cap + 1);
+    }
+
+    const T* gq = (const T*)bufs.queries + (size_t)q * di.dim;
+    for (int i = lane; i < di.dim; i += 64) qlds[i] = gq[i];
+
+    if (lane == 0) {
+        ss->ng_count = 0; ss->spt_count = 0; ss->dpq_len = 1; ss->checked = 0;
+        ss->oflow = 0; ss->terminate = 0; ss->break_flag = 0; ss->want_tree = 0;
+        ss->tree_checked = 0; ss->no_better = 0;
+        ng.a[0] = NodeDist{-1, MAXDIST};
+        spt.a[0] = NodeDist{-1, MAXDIST};
+        dpq[1] = MAXDIST;
+        for (int i = 0; i < cfg.k; i++) qrs[i] = QRes{-1, MAXDIST};
+    }
+    __syncthreads();
+
+    QCtx<T> c{&di, &cfg, qlds, dstage, istage, qrs, dpq, ng, spt,
+              bufs.visited + (size_t)q * cfg.vcap, (uint32_t)(cfg.vcap - 1),
+              ss, lane};
+
+    /* InitSearchTrees (KDTree.h:213): every tree root at bound 0 */
+    for (int t = 0; t < di.ntrees; t++)
+        kdt_search_node_dev<T, DM>(c, di.tree_start[t], 0.0f);
+    kdt_search_trees_dev<T, DM>(c, cfg.init_pivots);
+
+    const int deg = di.deg;
+    int popped = 0;
+
+    for (;;) {
+        __syncthreads();
+        if (ss->ng_count <= 0 || ss->terminate || ss->oflow) break;
+        if (lane == 0) ss->popped = ndheap_pop(c.ng, &ss->ng_count);
+        __syncthreads();
+        popped++;
+        NodeDist gnode = ss->popped;
+        const int32_t* row = di.graph + (size_t)gnode.node * deg;
+        int32_t nn = lane < deg ? row[lane] : -1;
+        uint64_t negm = __ballot(lane >= deg || nn < 0);
+        int firstneg = negm ? (int)__builtin_ctzll(negm) : 64;
+
+        if (lane == 0) {
+            /* KDTIndex.cpp:201-209: result + budget termination */
+            if (not_deleted(di, gnode.node)) {
+                if (!qrs_add(qrs, cfg.k, gnode.node, gnode.distance) &&
+                    ss->checked > cfg.max_check)
+                    ss->terminate = 1;
+            }
+            float worst = qrs[0].dist;
+            ss->fbcast = worst > gnode.distance ? worst : gnode.distance;
+        }
+        __syncthreads();
+        if (ss->terminate) break;
+        float upper_bound = ss->fbcast;
+
+        int already = 1;
+        if (lane < firstneg)
+            already = visited_test_insert(c.vtab, c.vmask, nn, &ss->oflow);
+        uint64_t candm = __ballot(lane < firstneg && !already);
+        int ncand = __popcll(candm);
+        if ((candm >> lane) & 1) {
+            int pos = __popcll(candm & ((1ull << lane) - 1));
+            istage[pos] = nn;
+        }
+        __syncthreads();
+        stage_dists<T, DM>(c, ncand);
+        if (lane == 0) {
+            /* KDTIndex.cpp:211-233: every computed neighbor joins the
+             * frontier; track whether any beat the upper bound */
+            int local_opt = 1;
+            for (int r = 0; r < ncand; r++) {
+                float dv = dstage[r];
+                if (dv <= upper_bound) local_opt = 0;
+                ss->checked++;
+                ndheap_insert(c.ng, &ss->ng_count, NodeDist{istage[r], dv}, &ss->oflow);
+            }
+            if (local_opt) ss->no_better++;
+            else ss->no_better = 0;
+            ss->want_tree = 0;
+            ss->break_flag = 0;
+            if (ss->no_better > cfg.nobetter_threshold) {
+                if (ss->tree_checked <= ss->checked / 10)
+                    ss->want_tree = 1;
+                else if (gnode.distance > qrs[0].dist)
+                    ss->break_flag = 1;
+            }
+        }
+        __syncthreads();
+        if (ss->break_flag) break;
+        if (ss->want_tree)
+            kdt_search_trees_dev<T, DM>(c, cfg.other_pivots + ss->checked);
+    }
+    __syncthreads();
+
+    if (lane == 0) {
+        bufs.oflow[q] = ss->oflow;
+        if (bufs.stats) {
+            bufs.stats[(size_t)q * 2 + 0] = ss->checked;
+            bufs.stats[(size_t)q * 2 + 1] = popped;
+        }
+        qrs_sort(qrs, cfg.k);
+    }
+    __syncthreads();
+    if (lane < cfg.k) {
+        bufs.out_vids[(size_t)q * cfg.k + lane] = qrs[lane].vid;
+        bufs.out_dists[(size_t)q * cfg.k + lane] = qrs[lane].dist;
+    }
+}
+
+/* ------------------------------------------------------------------ *
  * brute-force truth kernel (TruthSet::GenerateTruth semantics,
  * TruthSet.h:163): exact top-k by (dist, vid) over all non-deleted rows.
  * One query per wave; test-scale only.
@@ -801,8 +1005,12 @@ static int launch_one(const DevIndex& di, const SearchCfg& cfg,
 {
     size_t lds = lds_bytes(di.dim, sizeof(T), cfg, LDSHEAP);
     dim3 grid(cfg.nq), block(64);
-    hipLaunchKernelGGL((bkt_search_kernel<T, DM, LDSHEAP>), grid, block, lds,
-                       stream, di, cfg, bufs);
+    if (di.algo == ALGO_KDT)
+        hipLaunchKernelGGL((kdt_search_kernel<T, DM, LDSHEAP>), grid, block, lds,
+                           stream, di, cfg, bufs);
+    else
+        hipLaunchKernelGGL((bkt_search_kernel<T, DM, LDSHEAP>), grid, block, lds,
+                           stream, di, cfg, bufs);
     return (int)hipGetLastError();
 }
 
