@@ -51,6 +51,16 @@ CONFIGS = {
     "bkt_100m_d100_i8_cos": dict(n=100_000_000, d=100, dtype="i8",
                                  metric="Cosine", nq=10_000, k=10,
                                  ncenters=16384, sigma=30.0),
+    # BASELINE.json configs[3] — KDT, embedding shape
+    "kdt_10m_d768_f32_cos": dict(n=10_000_000, d=768, dtype="f32",
+                                 metric="Cosine", nq=10_000, k=10,
+                                 ncenters=8192, sigma=28.0, algo="KDT",
+                                 cand=128),
+    # KDT validation scale
+    "kdt_1m_d768_f32_cos": dict(n=1_000_000, d=768, dtype="f32",
+                                metric="Cosine", nq=10_000, k=10,
+                                ncenters=4096, sigma=28.0, algo="KDT",
+                                cand=128),
     # int8 validation scale (same dtype/metric path as config #3)
     "bkt_10m_d100_i8_cos": dict(n=10_000_000, d=100, dtype="i8",
                                 metric="Cosine", nq=10_000, k=10,
@@ -244,13 +254,20 @@ def main():
     t0 = time.time()
     x_np = x.cpu().numpy()
     arrays = build_index_arrays(
-        x_np, cfg["metric"], ntrees=args.ntrees, refine_rounds=args.refine,
-        device=device, normalized=False, verbose=(rank == 0))
+        x_np, cfg["metric"], algo=cfg.get("algo", "BKT"),
+        cand=cfg.get("cand", 256), ntrees=args.ntrees,
+        refine_rounds=args.refine, device=device, normalized=False,
+        verbose=(rank == 0))
     log(rank, f"index built ({time.time()-t0:.1f}s)")
     t0 = time.time()
-    ix = sptag_amd.AnnIndex.FromArrays(
-        arrays["vectors"], arrays["tree_start"], arrays["tree_nodes"],
-        arrays["graph"], cfg["metric"], device=local_rank)
+    if cfg.get("algo") == "KDT":
+        ix = sptag_amd.AnnIndex.FromArraysKDT(
+            arrays["vectors"], arrays["tree_start"], arrays["tree_nodes"],
+            arrays["graph"], cfg["metric"], device=local_rank)
+    else:
+        ix = sptag_amd.AnnIndex.FromArrays(
+            arrays["vectors"], arrays["tree_start"], arrays["tree_nodes"],
+            arrays["graph"], cfg["metric"], device=local_rank)
     log(rank, f"index uploaded ({time.time()-t0:.1f}s)")
 
     # truth runs in the STORED vector space (cosine bases are normalized on

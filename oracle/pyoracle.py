@@ -30,6 +30,8 @@ def load_library():
         ctypes.c_int32, ctypes.c_int32, ctypes.c_int, ctypes.c_int,
         ctypes.c_void_p, ctypes.c_int32, ctypes.c_void_p, ctypes.c_int32,
         ctypes.c_void_p, ctypes.c_int32, ctypes.c_void_p, ctypes.c_void_p]
+    lib.orc_create_kdt_index.restype = ctypes.c_void_p
+    lib.orc_create_kdt_index.argtypes = lib.orc_create_index.argtypes
     lib.orc_free_index.argtypes = [ctypes.c_void_p]
     lib.orc_search.restype = ctypes.c_int32
     lib.orc_search.argtypes = [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int32,
@@ -80,6 +82,28 @@ class OrcIndex:
             vectors.ctypes.data_as(ctypes.c_void_p),
             len(tree_start), tree_start.ctypes.data_as(ctypes.c_void_p),
             tree_nodes.size // 3, tree_nodes.ctypes.data_as(ctypes.c_void_p),
+            graph.shape[1], graph.ctypes.data_as(ctypes.c_void_p), delp)
+        return cls(h)
+
+    @classmethod
+    def from_arrays_kdt(cls, vectors, tree_start, kdt_nodes, graph, distmethod,
+                        deleted=None):
+        lib = load_library()
+        vectors = np.ascontiguousarray(vectors)
+        vt = 0 if vectors.dtype == np.float32 else 1
+        dm = {"L2": 0, "Cosine": 1}.get(distmethod, distmethod)
+        tree_start = np.ascontiguousarray(tree_start, dtype=np.int32)
+        kdt_nodes = np.ascontiguousarray(kdt_nodes, dtype=np.int32)
+        graph = np.ascontiguousarray(graph, dtype=np.int32)
+        delp = None
+        if deleted is not None:
+            deleted = np.ascontiguousarray(deleted, dtype=np.uint8)
+            delp = deleted.ctypes.data_as(ctypes.c_void_p)
+        h = lib.orc_create_kdt_index(
+            vectors.shape[0], vectors.shape[1], vt, dm,
+            vectors.ctypes.data_as(ctypes.c_void_p),
+            len(tree_start), tree_start.ctypes.data_as(ctypes.c_void_p),
+            kdt_nodes.size // 4, kdt_nodes.ctypes.data_as(ctypes.c_void_p),
             graph.shape[1], graph.ctypes.data_as(ctypes.c_void_p), delp)
         return cls(h)
 

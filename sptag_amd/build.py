@@ -180,6 +180,146 @@ def build_bkt_tree(vectors, *, kmeans_k=32, leaf_size=32, big_cluster=2048,
 
 
 # ------------------------------------------------------------------ #
+# KDT tree (reference KDTree.h:84-118 BuildTrees / :286+ DivideTree:
+# split on a high-variance dim at the mean; leaves encode -(vecid+1))
+# ------------------------------------------------------------------ #
+
+def build_kdt_tree(vectors, *, ntrees=1, sample_dims=16, seed=2016,
+                   device=None, max_levels=64, verbose=False):
+    """Level-vectorized kd-tree build. Returns (tree_start int32[ntrees],
+    kdt_nodes int32[N,4]) — the 4th column holds the float split_value bit
+    pattern (KDTNode layout, KDTree.h:22). Split dim = max-variance among a
+    per-level random dim subset (reference: top-5 variance dims on a
+    sample); split value = segment mean."""
+    device = device or _dev()
+    n, d = vectors.shape
+    xf = torch.as_tensor(vectors, device=device)
+    if xf.dtype != torch.float32:
+        xf = xf.float()
+    gen = torch.Generator(device=device)
+    gen.manual_seed(seed + 31)
+
+    tree_start = []
+    all_nodes = []
+    base = 0
+    for t in range(ntrees):
+        tree_start.append(base)
+        # node storage (numpy, grown per level)
+        left_np = np.zeros(1, dtype=np.int64)
+        right_np = np.zeros(1, dtype=np.int64)
+        sdim_np_all = np.zeros(1, dtype=np.int64)
+        sval_np_all = np.zeros(1, dtype=np.float64)
+        perm = torch.randperm(n, generator=gen, device=device)
+        seg = torch.zeros(n, dtype=torch.int64, device=device)
+        seg_node = np.array([0], dtype=np.int64)   # node idx per segment
+        nseg = 1
+        for level in range(max_levels):
+            cnt = torch.bincount(seg, minlength=nseg)
+            if int(cnt.max()) <= 1:
+                break
+            k16 = min(sample_dims, d)
+            dims = torch.randperm(d, generator=gen, device=device)[:k16]
+            xs = xf[perm][:, dims]                       # [n, k16]
+            sums = torch.zeros(nseg, k16, device=device).index_add_(0, seg, xs)
+            sqs = torch.zeros(nseg, k16, device=device).index_add_(0, seg, xs * xs)
+            cf = cnt.float().clamp(min=1)[:, None]
+            var = sqs / cf - (sums / cf) ** 2
+            sd_local = var.argmax(1)                     # [nseg]
+            split_dim = dims[sd_local]
+            split_val = sums.gather(1, sd_local[:, None]).squeeze(1) / cf.squeeze(1)
+            val_e = xs.gather(1, sd_local[seg][:, None]).squeeze(1)
+            side = val_e >= split_val[seg]               # right side (diff >= 0)
+            # guard degenerate splits (all equal): rank-split those segments
+            lcnt = torch.bincount(seg[~side], minlength=nseg)
+            rcnt = cnt - lcnt
+            bad = ((lcnt == 0) | (rcnt == 0)) & (cnt >= 2)
+            if bool(bad.any()):
+                starts = torch.cumsum(cnt, 0) - cnt
+                pos = torch.arange(n, device=device) - starts[seg]
+                rank_side = pos >= (cnt[seg] // 2)
+                side = torch.where(bad[seg], rank_side, side)
+                lcnt = torch.bincount(seg[~side], minlength=nseg)
+                rcnt = cnt - lcnt
+            # order members: (seg, side) stable
+            o1 = side.long().argsort(stable=True)
+            o2 = seg[o1].argsort(stable=True)
+            order = o1[o2]
+            perm = perm[order]
+            side = side[order]
+            # record this level's nodes (only segments with cnt >= 2 are
+            # internal; they were allocated a node idx in seg_node)
+            cnt_np = cnt.cpu().numpy()
+            l_np = lcnt.cpu().numpy()
+            r_np = rcnt.cpu().numpy()
+            sd_np = split_dim.cpu().numpy()
+            sv_np = split_val.cpu().numpy()
+            starts_np = np.concatenate([[0], np.cumsum(cnt_np)[:-1]])
+            perm_np = perm.cpu().numpy()
+            # children allocation
+            internal = cnt_np >= 2
+            child_internal_l = internal & (l_np >= 2)
+            child_internal_r = internal & (r_np >= 2)
+            n_new = int(child_internal_l.sum() + child_internal_r.sum())
+            next_idx = len(left_np)
+            new_ids_l = np.full(nseg, -1, dtype=np.int64)
+            new_ids_r = np.full(nseg, -1, dtype=np.int64)
+            alloc = np.cumsum(np.concatenate(
+                [child_internal_l.astype(np.int64), child_internal_r.astype(np.int64)]))
+            new_ids_l[child_internal_l] = next_idx + alloc[:nseg][child_internal_l] - 1
+            new_ids_r[child_internal_r] = next_idx + alloc[nseg:][child_internal_r] - 1
+            left_np = np.concatenate([left_np, np.zeros(n_new, dtype=np.int64)])
+            right_np = np.concatenate([right_np, np.zeros(n_new, dtype=np.int64)])
+            sdim_np_all = np.concatenate([sdim_np_all, np.zeros(n_new, dtype=np.int64)])
+            sval_np_all = np.concatenate([sval_np_all, np.zeros(n_new)])
+            # children: internal -> allocated idx, single member -> -(id+1),
+            # empty side -> out-of-range leaf (ignored by search)
+            lchild = np.full(nseg, -(np.int64(n) + 1), dtype=np.int64)
+            rchild = np.full(nseg, -(np.int64(n) + 1), dtype=np.int64)
+            lchild[child_internal_l] = new_ids_l[child_internal_l]
+            rchild[child_internal_r] = new_ids_r[child_internal_r]
+            leaf_l = internal & (l_np == 1)
+            leaf_r = internal & (r_np == 1)
+            lchild[leaf_l] = -(perm_np[starts_np[leaf_l]] + 1)
+            rchild[leaf_r] = -(perm_np[starts_np[leaf_r] + l_np[leaf_r]] + 1)
+            li = np.where(internal)[0]
+            tgt = seg_node[li]
+            left_np[tgt] = lchild[li]
+            right_np[tgt] = rchild[li]
+            sdim_np_all[tgt] = sd_np[li]
+            sval_np_all[tgt] = sv_np[li]
+            # next level: keep only members of internal children
+            seg2 = seg * 2 + side.long()
+            keep_seg = torch.zeros(nseg * 2, dtype=torch.bool, device=device)
+            node_of = torch.full((nseg * 2,), -1, dtype=torch.int64, device=device)
+            kl = torch.as_tensor(child_internal_l, device=device)
+            kr = torch.as_tensor(child_internal_r, device=device)
+            keep_seg[0::2] = kl
+            keep_seg[1::2] = kr
+            node_of[0::2] = torch.as_tensor(new_ids_l, device=device)
+            node_of[1::2] = torch.as_tensor(new_ids_r, device=device)
+            keep = keep_seg[seg2]
+            perm = perm[keep]
+            seg2 = seg2[keep]
+            uniq, seg = torch.unique(seg2, return_inverse=True)
+            seg_node = node_of[uniq].cpu().numpy()
+            nseg = uniq.numel()
+            n = n  # unchanged (total vector count for leaf encoding)
+            if nseg == 0:
+                break
+            if verbose and level % 8 == 0:
+                print(f"  kdt tree {t}: level {level}, {nseg} open segments")
+        nodes = np.zeros((len(left_np), 4), dtype=np.int32)
+        nodes[:, 0] = left_np + np.where(left_np >= 0, base, 0)
+        nodes[:, 1] = right_np + np.where(right_np >= 0, base, 0)
+        nodes[:, 2] = sdim_np_all
+        nodes[:, 3] = sval_np_all.astype(np.float32).view(np.int32)
+        all_nodes.append(nodes)
+        base += len(left_np)
+    return (np.array(tree_start, dtype=np.int32),
+            np.concatenate(all_nodes).astype(np.int32))
+
+
+# ------------------------------------------------------------------ #
 # TP-tree KNN candidates + RNG prune
 # ------------------------------------------------------------------ #
 
@@ -452,20 +592,24 @@ def _exact_l2(xf, q_rows, c_ids):
     return d.masked_fill(c_ids < 0, float("inf"))
 
 
-def build_index_arrays(vectors, distmethod, *, degree=32, ntrees=4,
+def build_index_arrays(vectors, distmethod, *, algo="BKT", degree=32, ntrees=4,
                        tpt_leaf=1000, cand=256, kmeans_k=32, leaf_size=32,
-                       refine_rounds=0, seed=2016, device=None,
+                       refine_rounds=0, kdt_trees=1, seed=2016, device=None,
                        normalized=False, verbose=False):
     """Full build: returns dict(vectors, tree_start, tree_nodes, graph) ready
     for AnnIndex.FromArrays (vectors already cosine-normalized when needed,
     as the reference stores them on disk)."""
     if not normalized:
         vectors = normalize_base(vectors, distmethod)
-    tree_start, tree_nodes = build_bkt_tree(
-        vectors, kmeans_k=kmeans_k, leaf_size=leaf_size, seed=seed,
-        device=device, verbose=verbose)
+    if algo == "KDT":
+        tree_start, tree_nodes = build_kdt_tree(
+            vectors, ntrees=kdt_trees, seed=seed, device=device, verbose=verbose)
+    else:
+        tree_start, tree_nodes = build_bkt_tree(
+            vectors, kmeans_k=kmeans_k, leaf_size=leaf_size, seed=seed,
+            device=device, verbose=verbose)
     if verbose:
-        print(f"  bkt tree: {len(tree_nodes)} nodes")
+        print(f"  {algo} tree: {len(tree_nodes)} nodes")
     graph, cids, cdst = build_rng_graph(
         vectors, degree=degree, ntrees=ntrees, tpt_leaf=tpt_leaf, cand=cand,
         seed=seed, device=device, verbose=verbose)
@@ -477,4 +621,4 @@ def build_index_arrays(vectors, distmethod, *, degree=32, ntrees=4,
         graph = graph.cpu().numpy()
     return {"vectors": vectors, "tree_start": tree_start,
             "tree_nodes": tree_nodes, "graph": graph,
-            "distmethod": distmethod}
+            "distmethod": distmethod, "algo": algo}

@@ -78,3 +78,25 @@ def test_normalize_int8_matches_reference_truncation():
     out = normalize_base(x, "Cosine")
     norms = np.linalg.norm(out.astype(np.float64), axis=1)
     assert (np.abs(norms - 127) < 8).all()
+
+
+def test_kdt_build_and_search_recall():
+    from sptag_amd.build import build_kdt_tree
+    from oracle.pyoracle import OrcIndex
+    n, d = 20000, 32
+    x, q = make_clustered(n, d, 200, nq=200)
+    arrays = build_index_arrays(x, "L2", algo="KDT", device="cpu", ntrees=4,
+                                tpt_leaf=500, cand=128)
+    kn = arrays["tree_nodes"]
+    assert kn.shape[1] == 4
+    # every vector appears exactly once as a leaf
+    leaves = np.concatenate([kn[:, 0], kn[:, 1]])
+    leaves = -leaves[leaves < 0] - 1
+    leaves = leaves[leaves < n]
+    assert len(np.unique(leaves)) == n
+    ix = OrcIndex.from_arrays_kdt(x, arrays["tree_start"], kn,
+                                  arrays["graph"], "L2")
+    tv, _ = ix.truth(q, 10, nthreads=4)
+    vids, _ = ix.search_batch(q, 10, 2048, nthreads=4)
+    hits = sum(len(set(tv[i]).intersection(vids[i])) for i in range(len(q)))
+    assert hits / (len(q) * 10) > 0.9, hits / (len(q) * 10)
