@@ -94,9 +94,13 @@ def gen_data(cfg, shard, world, device, torch):
         centers = torch.randn((cfg["ncenters"], d), generator=gen,
                               device=device) * cfg["sigma"]
     else:
-        # SIFT-shaped: overlapping mixture over the [0,255] box
-        centers = torch.rand((cfg["ncenters"], d), generator=gen,
-                             device=device) * 255.0
+        # SIFT-shaped: hierarchical mixture over the [0,255] box (super-
+        # centers -> centers -> points; BASELINE.md clusterability note)
+        supers = torch.rand((256, d), generator=gen, device=device) * 255.0
+        slab = torch.randint(0, 256, (cfg["ncenters"],), generator=gen,
+                             device=device)
+        centers = supers[slab] + torch.randn((cfg["ncenters"], d), generator=gen,
+                                             device=device) * (cfg["sigma"] * 1.5)
     lo = n * shard // world
     hi = n * (shard + 1) // world
     gen.manual_seed(2016 + 1 + shard)
