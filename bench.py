@@ -270,7 +270,8 @@ def main():
         refine_rounds=args.refine, device=device, normalized=False,
         verbose=(rank == 0))
     log(rank, f"index built ({time.time()-t0:.1f}s)")
-    t0 = time.time()
+    torch.cuda.empty_cache()   # release build-phase cache so the extension's
+    t0 = time.time()           # hipMalloc can place the index blobs
     if cfg.get("algo") == "KDT":
         ix = sptag_amd.AnnIndex.FromArraysKDT(
             arrays["vectors"], arrays["tree_start"], arrays["tree_nodes"],

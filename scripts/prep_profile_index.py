@@ -22,6 +22,7 @@ def main():
     x, q, lo = bench.gen_data(cfg, 0, 1, "cuda:0", torch)
     arrays = build_index_arrays(x.cpu().numpy(), cfg["metric"], ntrees=4,
                                 refine_rounds=0, device="cuda:0")
+    torch.cuda.empty_cache()
     ix = sptag_amd.AnnIndex.FromArrays(
         arrays["vectors"], arrays["tree_start"], arrays["tree_nodes"],
         arrays["graph"], cfg["metric"])
