@@ -50,7 +50,8 @@ CONFIGS = {
     # BASELINE.json configs[2] — int8 cosine SPACEV shape
     "bkt_100m_d100_i8_cos": dict(n=100_000_000, d=100, dtype="i8",
                                  metric="Cosine", nq=10_000, k=10,
-                                 ncenters=16384, sigma=30.0),
+                                 ncenters=16384, sigma=30.0, cand=96,
+                                 ntrees=2),
     # BASELINE.json configs[3] — KDT, embedding shape. Note: the KDT
     # algorithm's no-better-propagation termination caps recall on this
     # data family at ~0.93-0.94 for the REFERENCE implementation as well
@@ -266,7 +267,8 @@ def main():
     x_np = x.cpu().numpy()
     arrays = build_index_arrays(
         x_np, cfg["metric"], algo=cfg.get("algo", "BKT"),
-        cand=cfg.get("cand", 256), kdt_trees=2, ntrees=args.ntrees,
+        cand=cfg.get("cand", 256), kdt_trees=2,
+        ntrees=cfg.get("ntrees", args.ntrees),
         refine_rounds=args.refine, device=device, normalized=False,
         verbose=(rank == 0))
     log(rank, f"index built ({time.time()-t0:.1f}s)")

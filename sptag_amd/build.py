@@ -43,6 +43,19 @@ def normalize_base(vectors, distmethod):
 # BKT tree
 # ------------------------------------------------------------------ #
 
+def _ragged_arange(counts):
+    """[0..c0), [0..c1), ... concatenated (numpy)."""
+    counts = np.asarray(counts, dtype=np.int64)
+    total = int(counts.sum())
+    if total == 0:
+        return np.zeros(0, dtype=np.int64)
+    out = np.ones(total, dtype=np.int64)
+    out[0] = 0
+    starts = np.cumsum(counts)[:-1]
+    out[starts] = 1 - counts[:-1]
+    return np.cumsum(out)
+
+
 def _kmeans_split(xf, members, K, iters=8, sample=4096, gen=None):
     """k-means split of one cluster. Returns (groups: list of int64 tensors,
     reps: list of representative vector ids)."""
@@ -95,7 +108,7 @@ def _kmeans_split(xf, members, K, iters=8, sample=4096, gen=None):
     return groups, reps
 
 
-def build_bkt_tree(vectors, *, kmeans_k=32, leaf_size=32, big_cluster=2048,
+def build_bkt_tree(vectors, *, kmeans_k=32, leaf_size=32, big_cluster=4096,
                    seed=2016, device=None, verbose=False):
     """Returns (tree_start int32[ntrees], tree_nodes int32[N,3]) in the
     reference layout (BKTree.h:25 BKTNode {centerid,childStart,childEnd},
@@ -134,37 +147,75 @@ def build_bkt_tree(vectors, *, kmeans_k=32, leaf_size=32, big_cluster=2048,
 
     # vectorized bottom: each small cluster -> groups of <= leaf_size members
     # (ordered by distance to the cluster centroid), each group an internal
-    # node whose children are member leaves.
+    # node whose children are member leaves. Fully vectorized (no per-cluster
+    # python loop) so 100M-scale builds stay fast.
     if small:
         nbase = len(cent)
-        sizes = np.array([g.numel() for _, g in small])
+        dev = xf.device
+        members_all = torch.cat([g for _, g in small])
+        cl_of = torch.repeat_interleave(
+            torch.arange(len(small), device=dev),
+            torch.tensor([g.numel() for _, g in small], device=dev))
+        ncl = len(small)
+        sizes_t = torch.tensor([g.numel() for _, g in small], device=dev)
+        # order members within each cluster by distance to its centroid
+        pts = xf[members_all]
+        cen = torch.zeros((ncl, pts.shape[1]), device=dev)
+        cen.index_add_(0, cl_of, pts)
+        cen = cen / sizes_t[:, None].float()
+        dcent = ((pts - cen[cl_of]) ** 2).sum(1)
+        o1 = dcent.argsort()
+        o2 = cl_of[o1].argsort(stable=True)
+        order = o1[o2]
+        members_all = members_all[order]          # cluster-major, dist-ordered
+        mem_np = members_all.cpu().numpy()
+        sizes = sizes_t.cpu().numpy()
+        node_np = np.array([nd for nd, _ in small])
+
         gcnts = np.ceil(sizes / leaf_size).astype(np.int64)
-        extra = np.where(sizes > leaf_size, gcnts, 0) + sizes
+        has_groups = sizes > leaf_size
+        n_group_nodes = np.where(has_groups, gcnts, 0)
+        extra = n_group_nodes + sizes              # nodes per cluster
         offs = nbase + np.concatenate([[0], np.cumsum(extra)[:-1]])
         total = int(extra.sum())
         cent2 = np.empty(total, dtype=np.int64)
         cs2 = np.full(total, -1, dtype=np.int64)
         ce2 = np.full(total, -1, dtype=np.int64)
-        for (node, g), size, gcnt, off in zip(small, sizes, gcnts, offs):
-            xg = xf[g]
-            order = torch.cdist(xg, xg.mean(0, keepdim=True)).squeeze(1).argsort()
-            gi = g[order].cpu().numpy()
-            p = off - nbase
-            if size <= leaf_size:
-                cs[node] = off
-                ce[node] = off + size
-                cent2[p:p + size] = gi
-            else:
-                cs[node] = off
-                ce[node] = off + gcnt
-                lp = p + gcnt              # local cursor for leaf blocks
-                for gidx in range(int(gcnt)):
-                    mem = gi[gidx * leaf_size:(gidx + 1) * leaf_size]
-                    cent2[p + gidx] = mem[0]
-                    cs2[p + gidx] = nbase + lp
-                    ce2[p + gidx] = nbase + lp + len(mem)
-                    cent2[lp:lp + len(mem)] = mem
-                    lp += len(mem)
+
+        mstart = np.concatenate([[0], np.cumsum(sizes)[:-1]])
+        # clusters WITHOUT group level: children = member leaves
+        nog = ~has_groups
+        cs[node_np[nog]] = offs[nog]
+        ce[node_np[nog]] = offs[nog] + sizes[nog]
+        # their leaf slots: positions offs..offs+size-1 (local p = offs-nbase)
+        if nog.any():
+            lp = offs[nog] - nbase
+            idx = np.repeat(lp, sizes[nog]) + _ragged_arange(sizes[nog])
+            src = mem_np[np.repeat(mstart[nog], sizes[nog]) + _ragged_arange(sizes[nog])]
+            cent2[idx] = src
+        # clusters WITH a group level
+        wg = has_groups
+        if wg.any():
+            cs[node_np[wg]] = offs[wg]
+            ce[node_np[wg]] = offs[wg] + gcnts[wg]
+            gtot = int(gcnts[wg].sum())
+            # per group: cluster idx, group idx within cluster
+            gcl = np.repeat(np.where(wg)[0], gcnts[wg])
+            ggi = _ragged_arange(gcnts[wg])
+            gsz = np.minimum(leaf_size, sizes[gcl] - ggi * leaf_size)
+            # leaf block offsets: per cluster leaves start at offs+gcnt
+            leaf_base = offs[gcl] + gcnts[gcl] + ggi * leaf_size
+            # careful: groups are leaf_size-packed so cumulative offset works
+            gmemstart = mstart[gcl] + ggi * leaf_size
+            gpos = offs[gcl] - nbase + ggi        # group node local position
+            cent2[gpos] = mem_np[gmemstart]
+            cs2[gpos] = leaf_base
+            ce2[gpos] = leaf_base + gsz
+            # leaves
+            lidx = np.repeat(leaf_base - nbase, gsz) + _ragged_arange(gsz)
+            lsrc = mem_np[np.repeat(gmemstart, gsz) + _ragged_arange(gsz)]
+            cent2[lidx] = lsrc
+            del gtot
         cent = np.concatenate([cent, cent2])
         cs = np.concatenate([cs, cs2])
         ce = np.concatenate([ce, ce2])
@@ -390,33 +441,43 @@ def _leaf_knn(xf, perm, bounds, kper, chunk_rows=1_000_000):
     return ids, dst
 
 
-def _merge_candidates(ids_a, dst_a, ids_b, dst_b, cand, self_ids):
-    """Merge two candidate lists per point: dedupe ids, keep `cand` nearest.
-    All id tensors int32; invalid = -1/inf."""
-    cid = torch.cat([ids_a, ids_b], dim=1)
-    cdd = torch.cat([dst_a, dst_b], dim=1)
-    # drop self and duplicates (keep nearest occurrence)
-    order = cdd.argsort(dim=1, stable=True)
-    cid = torch.gather(cid, 1, order)
-    cdd = torch.gather(cdd, 1, order)
-    sid, sorder = cid.sort(dim=1, stable=True)
-    dup_sorted = torch.zeros_like(sid, dtype=torch.bool)
-    dup_sorted[:, 1:] = sid[:, 1:] == sid[:, :-1]
-    dup = torch.zeros_like(dup_sorted)
-    dup.scatter_(1, sorder, dup_sorted)
-    invalid = dup | (cid < 0) | (cid == self_ids[:, None])
-    cdd = cdd.masked_fill(invalid, float("inf"))
-    order = cdd.argsort(dim=1, stable=True)[:, :cand]
-    cid = torch.gather(cid, 1, order)
-    cdd = torch.gather(cdd, 1, order)
-    if cid.shape[1] < cand:
-        padn = cand - cid.shape[1]
-        cid = torch.cat([cid, torch.full((cid.shape[0], padn), -1,
-                                         dtype=cid.dtype, device=cid.device)], 1)
-        cdd = torch.cat([cdd, torch.full((cdd.shape[0], padn), float("inf"),
-                                         device=cdd.device)], 1)
-    cid = cid.masked_fill(~torch.isfinite(cdd), -1)
-    return cid, cdd
+def _merge_candidates(ids_a, dst_a, ids_b, dst_b, cand, self_ids,
+                      row_chunk=None):
+    """Merge two candidate lists per point: dedupe ids, keep `cand` nearest,
+    pad to exactly `cand` columns. Row-chunked: the transient sort/gather
+    tensors are ~6x the chunk size, which at 100M rows would not fit."""
+    n = ids_a.shape[0]
+    if row_chunk is None:
+        width = ids_a.shape[1] + ids_b.shape[1] + 1
+        row_chunk = max(65_536, int(4e9 // (width * 24)))
+    out_i = torch.empty((n, cand), dtype=ids_a.dtype, device=ids_a.device)
+    out_d = torch.empty((n, cand), dtype=dst_a.dtype, device=dst_a.device)
+    for s in range(0, n, row_chunk):
+        e = min(n, s + row_chunk)
+        cid = torch.cat([ids_a[s:e], ids_b[s:e]], dim=1)
+        cdd = torch.cat([dst_a[s:e], dst_b[s:e]], dim=1)
+        order = cdd.argsort(dim=1, stable=True)
+        cid = torch.gather(cid, 1, order)
+        cdd = torch.gather(cdd, 1, order)
+        sid, sorder = cid.sort(dim=1, stable=True)
+        dup_sorted = torch.zeros_like(sid, dtype=torch.bool)
+        dup_sorted[:, 1:] = sid[:, 1:] == sid[:, :-1]
+        dup = torch.zeros_like(dup_sorted)
+        dup.scatter_(1, sorder, dup_sorted)
+        invalid = dup | (cid < 0) | (cid == self_ids[s:e, None])
+        cdd = cdd.masked_fill(invalid, float("inf"))
+        order = cdd.argsort(dim=1, stable=True)[:, :cand]
+        ci = torch.gather(cid, 1, order)
+        cd = torch.gather(cdd, 1, order)
+        if ci.shape[1] < cand:
+            padn = cand - ci.shape[1]
+            ci = torch.cat([ci, torch.full((ci.shape[0], padn), -1,
+                                           dtype=ci.dtype, device=ci.device)], 1)
+            cd = torch.cat([cd, torch.full((cd.shape[0], padn), float("inf"),
+                                           device=cd.device)], 1)
+        out_i[s:e] = ci.masked_fill(~torch.isfinite(cd), -1)
+        out_d[s:e] = cd
+    return out_i, out_d
 
 
 def _prune_chunk(cand, d):
