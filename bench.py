@@ -230,6 +230,8 @@ def cpu_baseline_leg(index_dir, q_np, mc, k, seconds_budget=25):
 
 
 def main():
+    # reduce allocator fragmentation at 100M+ build scale
+    os.environ.setdefault("PYTORCH_ALLOC_CONF", "expandable_segments:True")
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=5)

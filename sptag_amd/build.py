@@ -442,16 +442,22 @@ def _leaf_knn(xf, perm, bounds, kper, chunk_rows=1_000_000):
 
 
 def _merge_candidates(ids_a, dst_a, ids_b, dst_b, cand, self_ids,
-                      row_chunk=None):
+                      row_chunk=None, out=None):
     """Merge two candidate lists per point: dedupe ids, keep `cand` nearest,
     pad to exactly `cand` columns. Row-chunked: the transient sort/gather
-    tensors are ~6x the chunk size, which at 100M rows would not fit."""
+    tensors are ~6x the chunk size, which at 100M rows would not fit.
+    `out=(ids, dst)` writes in place (may alias ids_a/dst_a: each chunk is
+    fully read before it is overwritten) — at 100M rows the fresh output
+    allocation alone is ~80GB."""
     n = ids_a.shape[0]
     if row_chunk is None:
         width = ids_a.shape[1] + ids_b.shape[1] + 1
         row_chunk = max(65_536, int(4e9 // (width * 24)))
-    out_i = torch.empty((n, cand), dtype=ids_a.dtype, device=ids_a.device)
-    out_d = torch.empty((n, cand), dtype=dst_a.dtype, device=dst_a.device)
+    if out is not None and out[0].shape[1] == cand:
+        out_i, out_d = out
+    else:
+        out_i = torch.empty((n, cand), dtype=ids_a.dtype, device=ids_a.device)
+        out_d = torch.empty((n, cand), dtype=dst_a.dtype, device=dst_a.device)
     for s in range(0, n, row_chunk):
         e = min(n, s + row_chunk)
         cid = torch.cat([ids_a[s:e], ids_b[s:e]], dim=1)
@@ -508,7 +514,8 @@ def build_rng_graph(vectors, *, degree=32, ntrees=4, tpt_leaf=1000, cand=256,
             ids, dst = _merge_candidates(tids, tdst, tids[:, :0], tdst[:, :0],
                                          cand, self_ids)
         else:
-            ids, dst = _merge_candidates(ids, dst, tids, tdst, cand, self_ids)
+            ids, dst = _merge_candidates(ids, dst, tids, tdst, cand, self_ids,
+                                         out=(ids, dst))
         del tids, tdst
         last_partition = (perm, bounds)
         if verbose:
