@@ -240,6 +240,7 @@ def main():
     ap.add_argument("--mc", type=int, default=0, help="force MaxCheck (skip sweep)")
     ap.add_argument("--no-cpu-baseline", action="store_true")
     ap.add_argument("--ntrees", type=int, default=4)
+    ap.add_argument("--cand", type=int, default=0)
     ap.add_argument("--refine", type=int, default=0)
     args = ap.parse_args()
 
@@ -269,7 +270,7 @@ def main():
     x_np = x.cpu().numpy()
     arrays = build_index_arrays(
         x_np, cfg["metric"], algo=cfg.get("algo", "BKT"),
-        cand=cfg.get("cand", 256), kdt_trees=2,
+        cand=args.cand or cfg.get("cand", 256), kdt_trees=2,
         ntrees=cfg.get("ntrees", args.ntrees),
         refine_rounds=args.refine, device=device, normalized=False,
         verbose=(rank == 0))
