@@ -548,11 +548,23 @@ def build_rng_graph(vectors, *, degree=32, ntrees=4, tpt_leaf=1000, cand=256,
         cd = ((pts - centroids[:, None, :]) ** 2).sum(-1)
         cd = cd.masked_fill(~pmask, float("inf"))
         medoid = pad.gather(1, cd.argmin(1, keepdim=True)).squeeze(1)  # [L]
-        lc2 = torch.cdist(centroids, centroids)
-        lc2.fill_diagonal_(float("inf"))
-        nleaf = lc2.topk(bknn, dim=1, largest=False).indices  # [L, bknn]
-        bridge_ids = medoid[nleaf][leaf_of].int()             # [n, bknn]
-        del pad, pmask, pts, cd, lc2
+        # exponential-rank ladder of nearest leaves: ranks 1,2,4,...: short
+        # bridges link adjacent leaves, long ones cross cluster groups —
+        # scale-independent (at 100M a tight cluster spans many leaves, so
+        # nearest-8 bridges alone stay inside it and islands return).
+        ranks = [1, 2, 4, 8, 16, 32, 64, 128]
+        ranks = [r for r in ranks if r < L][:bknn]
+        kmax = max(ranks) + 1
+        nleaf = torch.empty((L, len(ranks)), dtype=torch.int64, device=device)
+        lchunk = max(1024, int(2e9 // (L * 4)))
+        for s0 in range(0, L, lchunk):
+            e0 = min(L, s0 + lchunk)
+            dc = torch.cdist(centroids[s0:e0], centroids)
+            idxs = dc.topk(kmax + 1, dim=1, largest=False).indices
+            # drop self (rank 0), take the ladder ranks
+            nleaf[s0:e0] = idxs[:, 1:][:, [r - 1 for r in ranks]]
+        bridge_ids = medoid[nleaf][leaf_of].int()             # [n, len(ranks)]
+        del pad, pmask, pts, cd
 
     if point_chunk is None:
         point_chunk = _prune_chunk(cand, xf.shape[1])
