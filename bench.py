@@ -241,11 +241,14 @@ def main():
     ap.add_argument("--no-cpu-baseline", action="store_true")
     ap.add_argument("--ntrees", type=int, default=4)
     ap.add_argument("--cand", type=int, default=0)
+    ap.add_argument("--ncenters", type=int, default=0)
     ap.add_argument("--refine", type=int, default=0)
     args = ap.parse_args()
 
     import torch
-    cfg = CONFIGS[args.workload]
+    cfg = dict(CONFIGS[args.workload])
+    if args.ncenters:
+        cfg["ncenters"] = args.ncenters
     world = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
     local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
