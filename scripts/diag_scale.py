@@ -21,11 +21,15 @@ from oracle.pyoracle import OrcIndex
 
 def main():
     n = int(sys.argv[1])
+    ntrees = int(sys.argv[2]) if len(sys.argv) > 2 else 2
+    refine = int(sys.argv[3]) if len(sys.argv) > 3 else 0
+    cand = int(sys.argv[4]) if len(sys.argv) > 4 else 96
     cfg = dict(bench.CONFIGS["bkt_100m_d100_i8_cos"])
     cfg["n"] = n
     x, q, lo = bench.gen_data(cfg, 0, 1, "cuda:0", torch)
     t0 = time.time()
-    arrays = build_index_arrays(x.cpu().numpy(), "Cosine", cand=96, ntrees=2,
+    arrays = build_index_arrays(x.cpu().numpy(), "Cosine", cand=cand,
+                                ntrees=ntrees, refine_rounds=refine,
                                 device="cuda:0")
     print(f"built {n} in {time.time()-t0:.0f}s", flush=True)
     torch.cuda.empty_cache()
@@ -45,6 +49,7 @@ def main():
         print(f"GPU    mc={mc}: recall={bench.recall_at_k(gv, tv, 10):.4f}",
               flush=True)
 
+    return  # oracle leg not needed once kernel==oracle is established
     oix = OrcIndex.from_arrays(arrays["vectors"], arrays["tree_start"],
                                arrays["tree_nodes"], arrays["graph"], "Cosine")
     for mc in [4096, 16384]:
