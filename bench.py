@@ -91,9 +91,14 @@ def log(rank, *a):
         print("[bench]", *a, file=sys.stderr, flush=True)
 
 
+GEN_CHUNK = 1_000_000
+
+
 def gen_data(cfg, shard, world, device, torch):
-    """Deterministic mixture; every rank generates its contiguous shard
-    rows [lo, hi) plus the shared query set."""
+    """Deterministic mixture. The dataset is defined GLOBALLY in fixed 1M-row
+    chunks with per-chunk seeds, so rank r's rows [lo, hi) are identical for
+    every world size — N=1 and N=8 runs search the same data (strong-scaling
+    comparability)."""
     n, d = cfg["n"], cfg["d"]
     gen = torch.Generator(device=device)
     gen.manual_seed(2016)
@@ -111,10 +116,21 @@ def gen_data(cfg, shard, world, device, torch):
                                              device=device) * (cfg["sigma"] * 1.5)
     lo = n * shard // world
     hi = n * (shard + 1) // world
-    gen.manual_seed(2016 + 1 + shard)
     noise = 1.0 if cfg.get("style") == "emb" else cfg["sigma"]
-    lab = torch.randint(0, cfg["ncenters"], (hi - lo,), generator=gen, device=device)
-    x = centers[lab] + torch.randn((hi - lo, d), generator=gen, device=device) * noise
+    parts = []
+    for c0 in range(lo // GEN_CHUNK, (hi + GEN_CHUNK - 1) // GEN_CHUNK):
+        cs_, ce_ = c0 * GEN_CHUNK, min((c0 + 1) * GEN_CHUNK, n)
+        gen.manual_seed(2016 + 100 + c0)
+        lab = torch.randint(0, cfg["ncenters"], (ce_ - cs_,), generator=gen,
+                            device=device)
+        xc = centers[lab] + torch.randn((ce_ - cs_, d), generator=gen,
+                                        device=device) * noise
+        a = max(lo, cs_) - cs_
+        b = min(hi, ce_) - cs_
+        parts.append(xc[a:b])
+        del lab, xc
+    x = torch.cat(parts) if len(parts) > 1 else parts[0]
+    del parts
     gen.manual_seed(2016 + 9999)
     qlab = torch.randint(0, cfg["ncenters"], (cfg["nq"],), generator=gen, device=device)
     q = centers[qlab] + torch.randn((cfg["nq"], d), generator=gen, device=device) * noise
