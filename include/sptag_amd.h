@@ -116,6 +116,20 @@ void sptag_amd_last_stats(SptagAmdIndex* idx, double* kernel_ms,
 int sptag_amd_truth(SptagAmdIndex* idx, const void* queries, int32_t nq,
                     int32_t k, int32_t* out_vids, float* out_dists);
 
+/* Iterative (streaming) search — mirrors the reference's ResultIterator /
+ * SearchIndexIterativeNext protocol (inc/Core/VectorIndex.h:43-49,
+ * src/Core/ResultIterator.cpp) over a BATCH of nq per-query iterators whose
+ * traversal state persists on the device between calls. Each next() call
+ * returns up to `batch` further results per query, sorted, with the
+ * per-query result count and the sticky relaxed-monotonicity flag. */
+typedef struct SptagAmdIterBatch SptagAmdIterBatch;
+SptagAmdIterBatch* sptag_amd_iter_create(SptagAmdIndex* idx, const void* queries,
+                                         int32_t nq, int32_t max_check);
+int sptag_amd_iter_next(SptagAmdIterBatch* it, int32_t batch,
+                        int32_t* out_vids, float* out_dists,
+                        int32_t* out_counts, int32_t* out_relaxed);
+void sptag_amd_iter_free(SptagAmdIterBatch* it);
+
 /* Write the index back out in the reference's byte format
  * (vectors/tree/graph/deletes + indexloader.ini). */
 int sptag_amd_save_index(SptagAmdIndex* idx, const char* folder);

@@ -41,6 +41,14 @@ def load_library():
         ctypes.c_int32, ctypes.c_int, ctypes.c_void_p, ctypes.c_void_p]
     lib.orc_truth.argtypes = lib.orc_search_batch.argtypes[:4] + [
         ctypes.c_int, ctypes.c_void_p, ctypes.c_void_p]
+    lib.orc_iter_create.restype = ctypes.c_void_p
+    lib.orc_iter_create.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
+                                    ctypes.c_int32]
+    lib.orc_iter_next.restype = ctypes.c_int32
+    lib.orc_iter_next.argtypes = [ctypes.c_void_p, ctypes.c_int32,
+                                  ctypes.c_void_p, ctypes.c_void_p,
+                                  ctypes.c_void_p]
+    lib.orc_iter_free.argtypes = [ctypes.c_void_p]
     lib.orc_distance.restype = ctypes.c_float
     lib.orc_distance.argtypes = [ctypes.c_int, ctypes.c_int, ctypes.c_void_p,
                                  ctypes.c_void_p, ctypes.c_int32]
@@ -158,6 +166,9 @@ class OrcIndex:
             dists.ctypes.data_as(ctypes.c_void_p))
         return vids, dists
 
+    def iterate(self, query):
+        return OrcIter(self, query)
+
     def distance(self, x, y):
         dtype = np.float32 if self.valuetype == 0 else np.int8
         x = np.ascontiguousarray(x, dtype=dtype)
@@ -182,3 +193,30 @@ def recall_at_k(vids, truth_vids, truth_dists, k):
         # handled by the caller where needed)
         hits += len(tset.intersection(vids[i, :k].tolist()))
     return hits / (nq * k)
+
+
+class OrcIter:
+    """reference ResultIterator restatement (see sptag_oracle.c)."""
+
+    def __init__(self, index, query, max_check=8192):
+        self._lib = index._lib
+        self._index = index
+        q = index._qarr(query)[0]
+        self._h = self._lib.orc_iter_create(
+            index._h, np.ascontiguousarray(q).ctypes.data_as(ctypes.c_void_p),
+            max_check)
+
+    def next(self, batch):
+        vids = np.empty(batch, dtype=np.int32)
+        dists = np.empty(batch, dtype=np.float32)
+        rel = ctypes.c_int32()
+        cnt = self._lib.orc_iter_next(self._h, batch,
+                                      vids.ctypes.data_as(ctypes.c_void_p),
+                                      dists.ctypes.data_as(ctypes.c_void_p),
+                                      ctypes.byref(rel))
+        return cnt, vids, dists, rel.value
+
+    def __del__(self):
+        if getattr(self, "_h", None):
+            self._lib.orc_iter_free(self._h)
+            self._h = None

@@ -650,6 +650,138 @@ static int32_t orc_search_kdt(const OrcIndex* ix, const void* q, int32_t k,
     return checked;
 }
 
+/* ------------------------------------------------------------------ *
+ * Iterative (streaming) search — exact restatement of
+ * BKT::Index<T>::SearchIterative (src/Core/BKT/BKTIndex.cpp:354-427) and
+ * the ResultIterator driving protocol (src/Core/ResultIterator.cpp:30-54 +
+ * BKTIndex.cpp:660-676 SearchIndexIterativeNext: ResetResult each call).
+ * ------------------------------------------------------------------ */
+
+struct OrcIter {
+    const OrcIndex* ix;
+    void* query;               /* copied */
+    SearchSpace sp;
+    int relaxed_mono;          /* m_relaxedMono (sticky; WorkSpace.h) */
+    int first;
+    int max_check;
+};
+
+OrcIter* orc_iter_create(const OrcIndex* ix, const void* query, int32_t max_check)
+{
+    if (ix->algo != 0) return NULL;   /* BKT only (as in the reference BKT path) */
+    OrcIter* it = (OrcIter*)calloc(1, sizeof(OrcIter));
+    it->ix = ix;
+    size_t qb = (size_t)ix->dim * ix->esz;
+    it->query = malloc(qb);
+    memcpy(it->query, query, qb);
+    it->max_check = max_check;
+    ndheap_init(&it->sp.ng, max_check * 30);
+    ndheap_init(&it->sp.spt, max_check * 10);
+    distpq_init(&it->sp.results, max_check / 16 > 64 ? max_check / 16 : 64);
+    vset_init(&it->sp.visited, max_check * 2 + 64 * 32);
+    it->sp.checked = 0;
+    it->sp.tree_checked = 0;
+    it->sp.max_check = max_check;
+    it->first = 1;
+    return it;
+}
+
+void orc_iter_free(OrcIter* it)
+{
+    if (!it) return;
+    free(it->query);
+    vset_destroy(&it->sp.visited);
+    distpq_destroy(&it->sp.results);
+    ndheap_destroy(&it->sp.spt);
+    ndheap_destroy(&it->sp.ng);
+    free(it);
+}
+
+/* One Next(batch) call. Returns resultCount; out arrays sized batch.
+ * relaxed_mono receives the sticky flag. */
+int32_t orc_iter_next(OrcIter* it, int32_t batch, int32_t* out_vids,
+                      float* out_dists, int32_t* relaxed_mono)
+{
+    const OrcIndex* ix = it->ix;
+    SearchSpace* sp = &it->sp;
+    const void* q = it->query;
+
+    /* SearchIndexIterativeNext: workSpace->ResetResult(m_iMaxCheck, batch)
+     * (WorkSpace.h:276-283): fresh m_Results, counters zeroed; queues,
+     * visited set and m_relaxedMono persist. */
+    int res_cap = it->max_check / 16 > batch ? it->max_check / 16 : batch;
+    sp->results.a[1] = ORC_MAXDIST;
+    sp->results.length = 1;
+    sp->results.count = res_cap;
+    sp->checked = 0;
+    sp->tree_checked = 0;
+
+    QRes* storage = (QRes*)malloc(sizeof(QRes) * (size_t)batch);
+    QResultSet query;
+    qrs_init(&query, storage, batch);
+
+    if (it->first) {
+        init_search_trees(ix, q, sp);
+        search_trees(ix, q, sp, ORC_INIT_PIVOTS);
+        it->first = 0;
+    }
+
+    int count = 0;
+    const int checkPos = ix->deg - 1;
+    while (sp->ng.count > 0) {
+        NodeDist gnode = ndheap_pop(&sp->ng);
+        int32_t tmpNode = gnode.node;
+        const int32_t* row = ix->graph + (size_t)tmpNode * ix->deg;
+        if (not_deleted(ix, tmpNode)) {
+            qrs_add(&query, tmpNode, gnode.distance);
+            count++;
+            if (gnode.distance > distpq_worst(&sp->results) ||
+                sp->checked > it->max_check)
+                it->relaxed_mono = 1;
+        }
+        int32_t checkNode = row[checkPos];
+        if (checkNode < -1) {
+            /* iterative duplicate chain (BKTIndex.cpp:387-405): duplicates
+             * enter the FRONTIER so they stream out as separate results */
+            const BktNode* tnode = &ix->tree[-2 - checkNode];
+            int32_t i = -tnode->childStart;
+            while (i < tnode->childEnd) {
+                tmpNode = ix->tree[i].centerid;
+                if (not_deleted(ix, tmpNode)) {
+                    float d = idx_dist(ix, q, tmpNode);
+                    if (!vset_check_and_set(&sp->visited, tmpNode)) {
+                        NodeDist nd = { tmpNode, d };
+                        ndheap_insert(&sp->ng, nd);
+                    }
+                }
+                i++;
+            }
+        }
+        for (int i = 0; i <= checkPos; i++) {
+            int32_t nn = row[i];
+            if (nn < 0) break;
+            if (vset_check_and_set(&sp->visited, nn)) continue;
+            float d = idx_dist(ix, q, nn);
+            sp->checked++;
+            NodeDist nd = { nn, d };
+            ndheap_insert(&sp->ng, nd);
+            distpq_insert(&sp->results, d);
+        }
+        if (ndheap_top(&sp->ng).distance > ndheap_top(&sp->spt).distance)
+            search_trees(ix, q, sp, ORC_OTHER_PIVOTS + sp->checked);
+        if (count >= batch) break;
+    }
+
+    qrs_sort(&query);
+    for (int i = 0; i < batch; i++) {
+        out_vids[i] = query.r[i].vid;
+        out_dists[i] = query.r[i].dist;
+    }
+    if (relaxed_mono) *relaxed_mono = it->relaxed_mono;
+    free(storage);
+    return count;
+}
+
 void orc_search_batch(const OrcIndex* ix, const void* queries, int32_t nq,
                       int32_t k, int32_t max_check, int nthreads,
                       int32_t* out_vids, float* out_dists)

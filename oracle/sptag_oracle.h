@@ -82,6 +82,15 @@ void orc_search_batch(const OrcIndex* idx, const void* queries, int32_t nq,
 void orc_truth(const OrcIndex* idx, const void* queries, int32_t nq, int32_t k,
                int nthreads, int32_t* out_vids, float* out_dists);
 
+/* Iterative (streaming) search — reference SearchIterative /
+ * ResultIterator (BKTIndex.cpp:354-427, ResultIterator.cpp). BKT only. */
+typedef struct OrcIter OrcIter;
+OrcIter* orc_iter_create(const OrcIndex* idx, const void* query,
+                         int32_t max_check);
+int32_t orc_iter_next(OrcIter* it, int32_t batch, int32_t* out_vids,
+                      float* out_dists, int32_t* relaxed_mono);
+void orc_iter_free(OrcIter* it);
+
 /* Distance between two raw vectors with the reference's summation order
  * (DistanceUtils.cpp AVX512 chunk/fold order for float; exact integer math
  * for int8). Exposed for unit tests. */

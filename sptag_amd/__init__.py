@@ -78,6 +78,14 @@ def load_library():
         fn = getattr(lib, "sptag_amd_" + f)
         fn.restype = ctypes.c_int32
         fn.argtypes = [ctypes.c_void_p]
+    lib.sptag_amd_iter_create.restype = ctypes.c_void_p
+    lib.sptag_amd_iter_create.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
+                                          ctypes.c_int32, ctypes.c_int32]
+    lib.sptag_amd_iter_next.restype = ctypes.c_int
+    lib.sptag_amd_iter_next.argtypes = [ctypes.c_void_p, ctypes.c_int32,
+                                        ctypes.c_void_p, ctypes.c_void_p,
+                                        ctypes.c_void_p, ctypes.c_void_p]
+    lib.sptag_amd_iter_free.argtypes = [ctypes.c_void_p]
     lib.sptag_amd_gpu_available.restype = ctypes.c_int
     lib.sptag_amd_build_info.restype = ctypes.c_char_p
     _LIB = lib
@@ -256,8 +264,58 @@ class AnnIndex:
             raise SptagAmdError(rc, "truth")
         return vids, dists
 
+    def Iterate(self, queries, max_check=0):
+        """Streaming search: mirrors the reference GetIterator/Next protocol
+        (ResultIterator) for a batch of queries with device-persistent
+        traversal state."""
+        return IterBatch(self, queries, max_check)
+
     def Save(self, folder):
         os.makedirs(folder, exist_ok=True)
         rc = self._lib.sptag_amd_save_index(self._h, str(folder).encode())
         if rc != 0:
             raise SptagAmdError(rc, "save_index")
+
+
+class IterBatch:
+    """Batch of per-query result iterators (reference ResultIterator
+    semantics; state persists on the GPU between Next calls)."""
+
+    def __init__(self, index, queries, max_check=0):
+        self._lib = index._lib
+        self._index = index   # keep the index alive
+        queries = np.ascontiguousarray(queries,
+                                       dtype=_np_dtype(index.valuetype))
+        if queries.ndim == 1:
+            queries = queries[None, :]
+        self.nq = queries.shape[0]
+        self._h = self._lib.sptag_amd_iter_create(
+            index._h, queries.ctypes.data_as(ctypes.c_void_p), self.nq,
+            max_check)
+        if not self._h:
+            raise SptagAmdError(-6, "iter_create failed (see stderr)")
+
+    def Next(self, batch):
+        """Returns (vids [nq,batch], dists, counts [nq], relaxed [nq]):
+        the next `counts[i]` nearest results of query i, sorted; vid=-1
+        padding; relaxed = the sticky relaxed-monotonicity flag."""
+        vids = np.empty((self.nq, batch), dtype=np.int32)
+        dists = np.empty((self.nq, batch), dtype=np.float32)
+        counts = np.empty(self.nq, dtype=np.int32)
+        relaxed = np.empty(self.nq, dtype=np.int32)
+        rc = self._lib.sptag_amd_iter_next(
+            self._h, batch, vids.ctypes.data_as(ctypes.c_void_p),
+            dists.ctypes.data_as(ctypes.c_void_p),
+            counts.ctypes.data_as(ctypes.c_void_p),
+            relaxed.ctypes.data_as(ctypes.c_void_p))
+        if rc != 0:
+            raise SptagAmdError(rc, "iter_next")
+        return vids, dists, counts, relaxed
+
+    def Close(self):
+        if getattr(self, "_h", None):
+            self._lib.sptag_amd_iter_free(self._h)
+            self._h = None
+
+    def __del__(self):
+        self.Close()

@@ -69,6 +69,30 @@ inline size_t lds_bytes(int dim, size_t esz, const SearchCfg& c, bool heaps_in_l
     return b;
 }
 
+/* persistent per-iterator device state (iterative search) */
+struct IterState {
+    int32_t ng_count, spt_count, checked, tree_checked;
+    int32_t relaxed, first, oflow, pad;
+};
+
+/* iterative-search launch buffers: persistent state + per-call outputs */
+struct IterBufs {
+    const void* queries;   /* nq*dim (device, owned by the iterator batch) */
+    void* gheap_ng;        /* nq * (ng_cap+1) * 8 */
+    void* gheap_spt;       /* nq * (spt_cap+1) * 8 */
+    float* dpq;            /* nq * (dpq_cap+1) */
+    int32_t* visited;      /* nq * vcap (zeroed at create) */
+    IterState* state;      /* nq */
+    int32_t* out_vids;     /* nq * batch */
+    float* out_dists;      /* nq * batch */
+    int32_t* out_counts;   /* nq */
+    int32_t* out_relaxed;  /* nq */
+};
+
+int launch_bkt_iter(int valuetype, int distmethod, const DevIndex& di,
+                    const SearchCfg& cfg, const IterBufs& bufs, int batch,
+                    void* stream);
+
 /* launchers implemented in kernels.hip; return hipError_t as int */
 int launch_bkt_search(int valuetype, int distmethod, bool heaps_in_lds,
                       const DevIndex& di, const SearchCfg& cfg,

@@ -605,6 +605,141 @@ void sptag_amd_last_stats(SptagAmdIndex* ix, double* kernel_ms,
     if (popped) *popped = ix->last_popped;
 }
 
+struct SptagAmdIterBatch {
+    SptagAmdIndex* ix;
+    int32_t nq, max_check, vcap, ng_cap, spt_cap, dpq_alloc;
+    void* d_q = nullptr;
+    void* d_ng = nullptr;
+    void* d_spt = nullptr;
+    float* d_dpq = nullptr;
+    int32_t* d_visited = nullptr;
+    IterState* d_state = nullptr;
+    int32_t* d_ov = nullptr;
+    float* d_od = nullptr;
+    int32_t* d_oc = nullptr;
+    int32_t* d_or = nullptr;
+    int32_t out_cap = 0;
+};
+
+SptagAmdIterBatch* sptag_amd_iter_create(SptagAmdIndex* ix, const void* queries,
+                                         int32_t nq, int32_t max_check)
+{
+    if (!ix || !queries || nq <= 0) return nullptr;
+    if (ix->algo != ALGO_BKT) {
+        fprintf(stderr, "sptag_amd: iterative search is BKT-only\n");
+        return nullptr;
+    }
+    if (!sptag_amd_gpu_available() || !ix->d_vectors) {
+        fprintf(stderr, "sptag_amd: iterator requires a HIP device\n");
+        return nullptr;
+    }
+    if (max_check <= 0) max_check = ix->default_maxcheck;
+    auto* it = new SptagAmdIterBatch();
+    it->ix = ix;
+    it->nq = nq;
+    it->max_check = max_check;
+    it->ng_cap = max_check * 30;    /* reference WorkSpace.h:265 capacities */
+    it->spt_cap = max_check * 10;
+    it->dpq_alloc = std::max(max_check / 16, (int)MAX_K);
+    it->vcap = (int32_t)next_pow2((uint32_t)std::max(4096, max_check * 4));
+    size_t qb = (size_t)nq * ix->dim * ix->esz();
+    bool ok = hipMalloc(&it->d_q, qb) == hipSuccess &&
+        hipMemcpy(it->d_q, queries, qb, hipMemcpyHostToDevice) == hipSuccess &&
+        hipMalloc(&it->d_ng, (size_t)nq * (it->ng_cap + 1) * 8) == hipSuccess &&
+        hipMalloc(&it->d_spt, (size_t)nq * (it->spt_cap + 1) * 8) == hipSuccess &&
+        hipMalloc(&it->d_dpq, (size_t)nq * (it->dpq_alloc + 1) * 4) == hipSuccess &&
+        hipMalloc(&it->d_visited, (size_t)nq * it->vcap * 4) == hipSuccess &&
+        hipMemset(it->d_visited, 0, (size_t)nq * it->vcap * 4) == hipSuccess &&
+        hipMalloc(&it->d_state, (size_t)nq * sizeof(IterState)) == hipSuccess &&
+        hipMalloc(&it->d_oc, (size_t)nq * 4) == hipSuccess &&
+        hipMalloc(&it->d_or, (size_t)nq * 4) == hipSuccess;
+    if (ok) {
+        std::vector<IterState> init(nq);
+        for (auto& st : init) { st = IterState{}; st.first = 1; }
+        ok = hipMemcpy(it->d_state, init.data(), (size_t)nq * sizeof(IterState),
+                       hipMemcpyHostToDevice) == hipSuccess;
+    }
+    if (!ok) {
+        sptag_amd_iter_free(it);
+        return nullptr;
+    }
+    return it;
+}
+
+void sptag_amd_iter_free(SptagAmdIterBatch* it)
+{
+    if (!it) return;
+    if (it->d_q) (void)hipFree(it->d_q);
+    if (it->d_ng) (void)hipFree(it->d_ng);
+    if (it->d_spt) (void)hipFree(it->d_spt);
+    if (it->d_dpq) (void)hipFree(it->d_dpq);
+    if (it->d_visited) (void)hipFree(it->d_visited);
+    if (it->d_state) (void)hipFree(it->d_state);
+    if (it->d_ov) (void)hipFree(it->d_ov);
+    if (it->d_od) (void)hipFree(it->d_od);
+    if (it->d_oc) (void)hipFree(it->d_oc);
+    if (it->d_or) (void)hipFree(it->d_or);
+    delete it;
+}
+
+int sptag_amd_iter_next(SptagAmdIterBatch* it, int32_t batch,
+                        int32_t* out_vids, float* out_dists,
+                        int32_t* out_counts, int32_t* out_relaxed)
+{
+    if (!it || batch <= 0 || batch > MAX_K) return SPTAG_AMD_ERR_PARAM;
+    SptagAmdIndex* ix = it->ix;
+    std::lock_guard<std::mutex> g(ix->lock);
+    HIP_OR_FAIL(hipSetDevice(ix->device), SPTAG_AMD_ERR_NOGPU);
+    if (batch > it->out_cap) {
+        if (it->d_ov) (void)hipFree(it->d_ov);
+        if (it->d_od) (void)hipFree(it->d_od);
+        it->d_ov = nullptr; it->d_od = nullptr; it->out_cap = 0;
+        if (hipMalloc(&it->d_ov, (size_t)it->nq * batch * 4) != hipSuccess ||
+            hipMalloc(&it->d_od, (size_t)it->nq * batch * 4) != hipSuccess)
+            return SPTAG_AMD_ERR_OOM;
+        it->out_cap = batch;
+    }
+    SearchCfg cfg;
+    cfg.nq = it->nq;
+    cfg.k = batch;
+    cfg.max_check = it->max_check;
+    cfg.init_pivots = ix->init_pivots;
+    cfg.other_pivots = ix->other_pivots;
+    cfg.nobetter_threshold = ix->nobetter_threshold;
+    cfg.dpq_cap = std::max(it->max_check / 16, batch);  /* ResetResult cap */
+    cfg.vcap = it->vcap;
+    cfg.ng_cap = it->ng_cap;
+    cfg.spt_cap = it->spt_cap;
+
+    IterBufs ib;
+    ib.queries = it->d_q;
+    ib.gheap_ng = it->d_ng;
+    ib.gheap_spt = it->d_spt;
+    ib.dpq = it->d_dpq;
+    ib.visited = it->d_visited;
+    ib.state = it->d_state;
+    ib.out_vids = it->d_ov;
+    ib.out_dists = it->d_od;
+    ib.out_counts = it->d_oc;
+    ib.out_relaxed = it->d_or;
+    int err = launch_bkt_iter(ix->vt, ix->dm, ix->dev(), cfg, ib, batch, nullptr);
+    if (err != 0) {
+        fprintf(stderr, "sptag_amd: iter launch failed %d\n", err);
+        return SPTAG_AMD_ERR_INTERNAL;
+    }
+    HIP_OR_FAIL(hipDeviceSynchronize(), SPTAG_AMD_ERR_NOGPU);
+    HIP_OR_FAIL(hipMemcpy(out_vids, it->d_ov, (size_t)it->nq * batch * 4,
+                          hipMemcpyDeviceToHost), SPTAG_AMD_ERR_NOGPU);
+    HIP_OR_FAIL(hipMemcpy(out_dists, it->d_od, (size_t)it->nq * batch * 4,
+                          hipMemcpyDeviceToHost), SPTAG_AMD_ERR_NOGPU);
+    HIP_OR_FAIL(hipMemcpy(out_counts, it->d_oc, (size_t)it->nq * 4,
+                          hipMemcpyDeviceToHost), SPTAG_AMD_ERR_NOGPU);
+    if (out_relaxed)
+        HIP_OR_FAIL(hipMemcpy(out_relaxed, it->d_or, (size_t)it->nq * 4,
+                              hipMemcpyDeviceToHost), SPTAG_AMD_ERR_NOGPU);
+    return SPTAG_AMD_OK;
+}
+
 int sptag_amd_truth(SptagAmdIndex* ix, const void* queries, int32_t nq,
                     int32_t k, int32_t* out_vids, float* out_dists)
 {

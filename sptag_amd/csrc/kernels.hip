@@ -688,6 +688,173 @@ void bkt_search_kernel(DevIndex di, SearchCfg cfg, SearchBufs bufs)
 }
 
 /* ------------------------------------------------------------------ *
+ * Iterative (streaming) BKT search — reference SearchIterative
+ * (src/Core/BKT/BKTIndex.cpp:354-427) driven per ResultIterator::Next
+ * (src/Core/ResultIterator.cpp) with per-call ResetResult
+ * (BKTIndex.cpp:663). One wave per iterator; ALL traversal state
+ * (frontier/tree heaps at the reference's own capacities, visited table,
+ * results queue, counters) persists in global scratch between calls.
+ * ------------------------------------------------------------------ */
+
+template <typename T, int DM>
+__global__ __launch_bounds__(64, SPTAG_LB_WAVES)
+void bkt_iter_kernel(DevIndex di, SearchCfg cfg, IterBufs ib, int batch)
+{
+    const int q = blockIdx.x;
+    if (q >= cfg.nq) return;
+    const int lane = threadIdx.x;
+
+    extern __shared__ char smem[];
+    size_t off = 0;
+    T* qlds = (T*)(smem + off);
+    off += ((size_t)di.dim * sizeof(T) + 15) & ~15ul;
+    float* dstage = (float*)(smem + off); off += MAX_DEG * 4;
+    int32_t* istage = (int32_t*)(smem + off); off += MAX_DEG * 4;
+    QRes* qrs = (QRes*)(smem + off); off += (size_t)batch * 8;
+    SerialState* ss = (SerialState*)(smem + off); off += 64;
+
+    float* dpq = ib.dpq + (size_t)q * (cfg.dpq_cap + 1);
+    HeapRef ng, spt;
+    ng.cap = cfg.ng_cap;  ng.ref_cap = cfg.ng_cap;   /* reference capacities */
+    spt.cap = cfg.spt_cap; spt.ref_cap = cfg.spt_cap;
+    ng.a = (NodeDist*)ib.gheap_ng + (size_t)q * (cfg.ng_cap + 1);
+    spt.a = (NodeDist*)ib.gheap_spt + (size_t)q * (cfg.spt_cap + 1);
+
+    const T* gq = (const T*)ib.queries + (size_t)q * di.dim;
+    for (int i = lane; i < di.dim; i += 64) qlds[i] = gq[i];
+
+    IterState st = ib.state[q];
+    if (lane == 0) {
+        ss->ng_count = st.ng_count;
+        ss->spt_count = st.spt_count;
+        ss->oflow = st.oflow;
+        /* ResetResult (WorkSpace.h:276): fresh m_Results + zeroed counters;
+         * queues, visited set and the sticky relaxed flag persist. */
+        ss->dpq_len = 1;
+        dpq[1] = MAXDIST;
+        ss->checked = 0;
+        ss->tree_checked = 0;
+        ss->terminate = 0; ss->break_flag = 0; ss->want_tree = 0;
+        ss->no_better = st.relaxed;      /* reuse slot: sticky relaxedMono */
+        for (int i = 0; i < batch; i++) qrs[i] = QRes{-1, MAXDIST};
+        if (st.first) {
+            ng.a[0] = NodeDist{-1, MAXDIST};
+            spt.a[0] = NodeDist{-1, MAXDIST};
+        }
+    }
+    __syncthreads();
+
+    QCtx<T> c{&di, &cfg, qlds, dstage, istage, qrs, dpq, ng, spt,
+              ib.visited + (size_t)q * cfg.vcap, (uint32_t)(cfg.vcap - 1),
+              ss, lane};
+
+    if (st.first) {
+        init_search_trees_dev<T, DM>(c);
+        __syncthreads();
+        search_trees_dev<T, DM>(c, cfg.init_pivots);
+    }
+
+    const int deg = di.deg;
+    const int checkPos = deg - 1;
+    int count = 0;
+
+    for (;;) {
+        __syncthreads();
+        if (ss->ng_count <= 0 || ss->oflow || count >= batch) break;
+        if (lane == 0) ss->popped = ndheap_pop(c.ng, &ss->ng_count);
+        __syncthreads();
+        NodeDist gnode = ss->popped;
+        const int32_t* row = di.graph + (size_t)gnode.node * deg;
+        int32_t nn = lane < deg ? row[lane] : -1;
+        uint64_t negm = __ballot(lane >= deg || nn < 0);
+        int firstneg = negm ? (int)__builtin_ctzll(negm) : 64;
+        int32_t checkNode = __shfl(nn, checkPos);
+
+        if (lane == 0) {
+            int cnt_new = count;
+            if (not_deleted(di, gnode.node)) {
+                qrs_add(qrs, batch, gnode.node, gnode.distance);
+                cnt_new = count + 1;
+                if (gnode.distance > dpq[1] || ss->checked > cfg.max_check)
+                    ss->no_better = 1;                   /* relaxedMono */
+            }
+            ss->fbcast = __int_as_float(cnt_new);
+        }
+        __syncthreads();
+        count = __float_as_int(ss->fbcast);
+
+        if (checkNode < -1) {
+            /* iterative duplicate chain (BKTIndex.cpp:387-405): duplicates
+             * enter the frontier; stage chunk distances, lane 0 inserts */
+            const int32_t* tn = &di.tree_nodes[(size_t)(-2 - checkNode) * 3];
+            int32_t cs0 = -tn[1], ce0 = tn[2];
+            for (int base = cs0; base < ce0; base += 64) {
+                int cnt = min(64, ce0 - base);
+                if (lane < cnt)
+                    istage[lane] = di.tree_nodes[(size_t)(base + lane) * 3];
+                __syncthreads();
+                stage_dists<T, DM>(c, cnt);
+                if (lane == 0) {
+                    for (int i = 0; i < cnt; i++) {
+                        int32_t tmpNode = istage[i];
+                        if (!not_deleted(di, tmpNode)) continue;
+                        if (!visited_test_insert(c.vtab, c.vmask, tmpNode, &ss->oflow))
+                            ndheap_insert(c.ng, &ss->ng_count,
+                                          NodeDist{tmpNode, dstage[i]}, &ss->oflow);
+                    }
+                }
+                __syncthreads();
+            }
+        }
+
+        int already = 1;
+        if (lane < firstneg)
+            already = visited_test_insert(c.vtab, c.vmask, nn, &ss->oflow);
+        uint64_t candm = __ballot(lane < firstneg && !already);
+        int ncand = __popcll(candm);
+        if ((candm >> lane) & 1) {
+            int pos = __popcll(candm & ((1ull << lane) - 1));
+            istage[pos] = nn;
+        }
+        __syncthreads();
+        stage_dists<T, DM>(c, ncand);
+        if (lane == 0) {
+            for (int r = 0; r < ncand; r++) {
+                float dv = dstage[r];
+                ss->checked++;
+                ndheap_insert(c.ng, &ss->ng_count, NodeDist{istage[r], dv}, &ss->oflow);
+                dpq_insert(dpq, &ss->dpq_len, cfg.dpq_cap, dv);
+            }
+            ss->want_tree = (ndheap_top(c.ng, ss->ng_count).distance >
+                             ndheap_top(c.spt, ss->spt_count).distance);
+        }
+        __syncthreads();
+        if (ss->want_tree)
+            search_trees_dev<T, DM>(c, cfg.other_pivots + ss->checked);
+    }
+    __syncthreads();
+
+    if (lane == 0) {
+        qrs_sort(qrs, batch);
+        st.ng_count = ss->ng_count;
+        st.spt_count = ss->spt_count;
+        st.checked = ss->checked;
+        st.tree_checked = ss->tree_checked;
+        st.relaxed = ss->no_better;
+        st.first = 0;
+        st.oflow = ss->oflow;
+        ib.state[q] = st;
+        ib.out_counts[q] = count;
+        ib.out_relaxed[q] = st.relaxed;
+    }
+    __syncthreads();
+    if (lane < batch) {
+        ib.out_vids[(size_t)q * batch + lane] = qrs[lane].vid;
+        ib.out_dists[(size_t)q * batch + lane] = qrs[lane].dist;
+    }
+}
+
+/* ------------------------------------------------------------------ *
  * KDT search (reference src/Core/KDT/KDTIndex.cpp:184-241 +
  * inc/Core/Common/KDTree.h:213-273). Shares the heaps/visited/top-k/
  * distance machinery; differs in the seed descent (kd-tree with
@@ -1041,6 +1208,28 @@ static int launch_truth_one(const DevIndex& di, const void* queries, int32_t nq,
     hipLaunchKernelGGL((truth_kernel<T, DM>), dim3(nq), dim3(64), lds, s,
                        di, queries, nq, k, ov, od);
     return (int)hipGetLastError();
+}
+
+template <typename T, int DM>
+static int launch_iter_one(const DevIndex& di, const SearchCfg& cfg,
+                           const IterBufs& ib, int batch, hipStream_t s)
+{
+    size_t lds = (((size_t)di.dim * sizeof(T) + 15) & ~15ul) + MAX_DEG * 8 +
+                 (size_t)batch * 8 + 64;
+    hipLaunchKernelGGL((bkt_iter_kernel<T, DM>), dim3(cfg.nq), dim3(64), lds, s,
+                       di, cfg, ib, batch);
+    return (int)hipGetLastError();
+}
+
+int launch_bkt_iter(int vt, int dm, const DevIndex& di, const SearchCfg& cfg,
+                    const IterBufs& ib, int batch, void* stream)
+{
+    hipStream_t s = (hipStream_t)stream;
+    if (vt == VT_FLOAT && dm == DM_L2) return launch_iter_one<float, DM_L2>(di, cfg, ib, batch, s);
+    if (vt == VT_FLOAT && dm == DM_COSINE) return launch_iter_one<float, DM_COSINE>(di, cfg, ib, batch, s);
+    if (vt == VT_INT8 && dm == DM_L2) return launch_iter_one<int8_t, DM_L2>(di, cfg, ib, batch, s);
+    if (vt == VT_INT8 && dm == DM_COSINE) return launch_iter_one<int8_t, DM_COSINE>(di, cfg, ib, batch, s);
+    return (int)hipErrorInvalidValue;
 }
 
 int launch_truth(int vt, int dm, const DevIndex& di, const void* queries,

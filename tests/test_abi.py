@@ -93,3 +93,56 @@ def test_reference_searcher_accepts_saved_folder(tmp_path):
     ref = g["results"][2048]
     np.testing.assert_array_equal(rec["vid"], ref["vid"])
     np.testing.assert_array_equal(rec["dist"], ref["dist"])
+
+
+def test_kdt_save_roundtrip(tmp_path):
+    """KDT index folders round-trip byte-identically and stay loadable by
+    the oracle (pinned to the reference loader)."""
+    g = load_golden("kdt_f32_l2_n10k_d32")
+    ix = AnnIndex.Load(g["index"])
+    out = tmp_path / "saved"
+    ix.Save(str(out))
+    for f in ["vectors.bin", "tree.bin", "graph.bin"]:
+        a = open(os.path.join(g["index"], f), "rb").read()
+        b = open(out / f, "rb").read()
+        assert a == b, f"{f} differs after KDT save"
+    from oracle.pyoracle import OrcIndex
+    oix = OrcIndex.load(str(out))
+    assert oix.n == ix.n
+
+
+@pytest.mark.skipif(
+    not os.path.exists(os.path.join(REPO, "oracle", "_ref", "indexsearcher")),
+    reason="reference binaries not built in this environment")
+def test_reference_searcher_accepts_saved_kdt_folder(tmp_path):
+    import subprocess
+    g = load_golden("kdt_f32_l2_n10k_d32")
+    ix = AnnIndex.Load(g["index"])
+    out = tmp_path / "saved"
+    ix.Save(str(out))
+    res = tmp_path / "res.bin"
+    subprocess.run(
+        [os.path.join(REPO, "oracle", "_ref", "indexsearcher"),
+         "-d", str(ix.dim), "-v", "Float", "-f", "DEFAULT",
+         "-i", os.path.join(g["dir"], "queries.bin"), "-x", str(out),
+         "-k", "10", "-m", "2048", "-t", "2", "-of", "1", "-o", str(res)],
+        check=True, capture_output=True, cwd=tmp_path)
+    raw = open(res, "rb").read()
+    rec = np.frombuffer(raw[8:], dtype=np.dtype([("vid", np.int32),
+                                                 ("dist", np.float32)]))
+    rec = rec.reshape(g["meta"]["nq"], 10)
+    ref = g["results"][2048]
+    np.testing.assert_array_equal(rec["vid"], ref["vid"])
+    np.testing.assert_array_equal(rec["dist"], ref["dist"])
+
+
+def test_error_paths(tmp_path):
+    """C-ABI failure modes are loud, not silent."""
+    import sptag_amd as sa
+    lib = sa.load_library()
+    assert lib.sptag_amd_load_index(str(tmp_path / "nonexistent").encode(), 0) is None
+    # unsupported algo in ini
+    d = tmp_path / "bad"
+    d.mkdir()
+    (d / "indexloader.ini").write_text("[Index]\nIndexAlgoType=SPANN\n")
+    assert lib.sptag_amd_load_index(str(d).encode(), 0) is None

@@ -1,0 +1,63 @@
+"""Iterative (streaming) search parity.
+
+Golden fixtures (iter_b8_c5.bin) were produced by driving the REFERENCE's
+own ResultIterator (oracle/_ref/iterprobe, linked against the reference
+objects): per query, 5 Next(8) calls recording count, the sticky
+relaxed-monotonicity flag, ids and distances. The oracle restatement and
+the GPU iterator must both reproduce them bit-exactly."""
+import os
+
+import numpy as np
+import pytest
+
+from conftest import GOLDEN, load_golden
+from oracle.pyoracle import OrcIndex
+
+FIXTURES = ["f32_l2_n10k_d32", "i8_l2_n10k_d100"]
+
+
+def load_iter_golden(name):
+    raw = open(os.path.join(GOLDEN, name, "iter_b8_c5.bin"), "rb").read()
+    nq, batch, ncalls = np.frombuffer(raw[:12], dtype=np.int32)
+    rec = np.frombuffer(raw[12:], dtype=np.int32).reshape(nq, ncalls,
+                                                          2 + 2 * batch)
+    return int(nq), int(batch), int(ncalls), rec
+
+
+@pytest.mark.parametrize("name", FIXTURES)
+def test_oracle_iterative_bit_exact(name):
+    g = load_golden(name)
+    nq, batch, ncalls, rec = load_iter_golden(name)
+    ix = OrcIndex.load(g["index"])
+    for i in range(nq):
+        it = ix.iterate(g["queries"][i])
+        for c in range(ncalls):
+            cnt, vids, dists, rel = it.next(batch)
+            gold = rec[i, c]
+            assert cnt == gold[0], (name, i, c)
+            assert rel == gold[1], (name, i, c)
+            np.testing.assert_array_equal(vids[:cnt], gold[2::2][:cnt])
+            np.testing.assert_array_equal(dists[:cnt],
+                                          gold[3::2].view(np.float32)[:cnt])
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("name", FIXTURES)
+def test_gpu_iterative_bit_exact(name):
+    from sptag_amd import AnnIndex
+    g = load_golden(name)
+    nq, batch, ncalls, rec = load_iter_golden(name)
+    ix = AnnIndex.Load(g["index"])
+    it = ix.Iterate(g["queries"], max_check=8192)
+    for c in range(ncalls):
+        vids, dists, counts, relaxed = it.Next(batch)
+        np.testing.assert_array_equal(counts, rec[:, c, 0],
+                                      err_msg=f"{name} call {c} counts")
+        np.testing.assert_array_equal(relaxed, rec[:, c, 1],
+                                      err_msg=f"{name} call {c} relaxed")
+        for i in range(nq):
+            cnt = counts[i]
+            np.testing.assert_array_equal(vids[i, :cnt], rec[i, c, 2::2][:cnt])
+            np.testing.assert_array_equal(
+                dists[i, :cnt], rec[i, c, 3::2].view(np.float32)[:cnt])
+    it.Close()
