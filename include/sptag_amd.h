@@ -130,6 +130,14 @@ int sptag_amd_iter_next(SptagAmdIterBatch* it, int32_t batch,
                         int32_t* out_counts, int32_t* out_relaxed);
 void sptag_amd_iter_free(SptagAmdIterBatch* it);
 
+/* Incremental delete — mirrors VectorIndex::DeleteIndex(SizeType)
+ * (inc/Core/VectorIndex.h, BKT DeleteIndex -> Labelset::Insert,
+ * src/Core/BKT/BKTIndex.cpp:896): flags the ids; subsequent searches
+ * filter them with the reference's CheckIfNotDeleted dispatch (already
+ * covered by the parity suite). Persisted by save_index in deletes.bin. */
+int sptag_amd_delete(SptagAmdIndex* idx, const int32_t* vids, int32_t n);
+int64_t sptag_amd_deleted_count(const SptagAmdIndex* idx);
+
 /* Write the index back out in the reference's byte format
  * (vectors/tree/graph/deletes + indexloader.ini). */
 int sptag_amd_save_index(SptagAmdIndex* idx, const char* folder);

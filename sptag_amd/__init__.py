@@ -78,6 +78,11 @@ def load_library():
         fn = getattr(lib, "sptag_amd_" + f)
         fn.restype = ctypes.c_int32
         fn.argtypes = [ctypes.c_void_p]
+    lib.sptag_amd_delete.restype = ctypes.c_int
+    lib.sptag_amd_delete.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
+                                     ctypes.c_int32]
+    lib.sptag_amd_deleted_count.restype = ctypes.c_int64
+    lib.sptag_amd_deleted_count.argtypes = [ctypes.c_void_p]
     lib.sptag_amd_iter_create.restype = ctypes.c_void_p
     lib.sptag_amd_iter_create.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
                                           ctypes.c_int32, ctypes.c_int32]
@@ -263,6 +268,19 @@ class AnnIndex:
         if rc != 0:
             raise SptagAmdError(rc, "truth")
         return vids, dists
+
+    def Delete(self, vids):
+        """Flag vector ids as deleted (reference DeleteIndex semantics);
+        searches exclude them from then on."""
+        vids = np.ascontiguousarray(vids, dtype=np.int32)
+        rc = self._lib.sptag_amd_delete(
+            self._h, vids.ctypes.data_as(ctypes.c_void_p), vids.size)
+        if rc != 0:
+            raise SptagAmdError(rc, "delete")
+
+    @property
+    def deleted_count(self):
+        return self._lib.sptag_amd_deleted_count(self._h)
 
     def Iterate(self, queries, max_check=0):
         """Streaming search: mirrors the reference GetIterator/Next protocol

@@ -767,6 +767,37 @@ int sptag_amd_truth(SptagAmdIndex* ix, const void* queries, int32_t nq,
     return rc;
 }
 
+int sptag_amd_delete(SptagAmdIndex* ix, const int32_t* vids, int32_t n)
+{
+    if (!ix || !vids || n <= 0) return SPTAG_AMD_ERR_PARAM;
+    std::lock_guard<std::mutex> g(ix->lock);
+    if (ix->h_deleted.empty()) ix->h_deleted.assign((size_t)ix->n, 0);
+    for (int32_t i = 0; i < n; i++) {
+        int32_t v = vids[i];
+        if (v < 0 || v >= ix->n) return SPTAG_AMD_ERR_PARAM;
+        if (!ix->h_deleted[v]) {
+            ix->h_deleted[v] = 1;
+            ix->deleted_count++;
+        }
+    }
+    ix->has_deleted = ix->deleted_count > 0;
+    if (sptag_amd_gpu_available() && ix->d_vectors) {
+        HIP_OR_FAIL(hipSetDevice(ix->device), SPTAG_AMD_ERR_NOGPU);
+        if (!ix->d_deleted)
+            HIP_OR_FAIL(hipMalloc(&ix->d_deleted, ix->h_deleted.size()),
+                        SPTAG_AMD_ERR_OOM);
+        HIP_OR_FAIL(hipMemcpy(ix->d_deleted, ix->h_deleted.data(),
+                              ix->h_deleted.size(), hipMemcpyHostToDevice),
+                    SPTAG_AMD_ERR_NOGPU);
+    }
+    return SPTAG_AMD_OK;
+}
+
+int64_t sptag_amd_deleted_count(const SptagAmdIndex* ix)
+{
+    return ix ? ix->deleted_count : -1;
+}
+
 int sptag_amd_save_index(SptagAmdIndex* ix, const char* folder)
 {
     if (!ix || !folder) return SPTAG_AMD_ERR_PARAM;
