@@ -130,6 +130,14 @@ int sptag_amd_iter_next(SptagAmdIterBatch* it, int32_t batch,
                         int32_t* out_counts, int32_t* out_relaxed);
 void sptag_amd_iter_free(SptagAmdIterBatch* it);
 
+/* Online add — mirrors BKT AddIndex below the tree-rebuild threshold
+ * (src/Core/BKT/BKTIndex.cpp:902-970): appends the vectors, then per new
+ * node a GPU refine search (AddCEF=500, MaxCheckForRefineGraph) followed by
+ * the reference's RNG rebuild + two-way neighbor inserts. Requires a HIP
+ * device (the refine searches run on it). */
+int sptag_amd_add(SptagAmdIndex* idx, const void* vectors, int32_t nadd,
+                  int normalized);
+
 /* Incremental delete — mirrors VectorIndex::DeleteIndex(SizeType)
  * (inc/Core/VectorIndex.h, BKT DeleteIndex -> Labelset::Insert,
  * src/Core/BKT/BKTIndex.cpp:896): flags the ids; subsequent searches

@@ -452,11 +452,26 @@ enum { ORC_INIT_PIVOTS = 50, ORC_OTHER_PIVOTS = 4 };
 static int32_t orc_search_kdt(const OrcIndex* ix, const void* q, int32_t k,
                               int32_t max_check, int32_t* out_vids, float* out_dists);
 
+static int32_t orc_search_bkt(const OrcIndex* ix, const void* q, int32_t k,
+                              int32_t max_check, int search_dup,
+                              int search_deleted, int32_t* out_vids,
+                              float* out_dists);
+
 int32_t orc_search(const OrcIndex* ix, const void* q, int32_t k,
                    int32_t max_check, int32_t* out_vids, float* out_dists)
 {
     if (ix->algo == 1)
         return orc_search_kdt(ix, q, k, max_check, out_vids, out_dists);
+    /* default entry: searchDeleted=false, searchDuplicated=true
+     * (BKTIndex.cpp:615) */
+    return orc_search_bkt(ix, q, k, max_check, 1, 0, out_vids, out_dists);
+}
+
+static int32_t orc_search_bkt(const OrcIndex* ix, const void* q, int32_t k,
+                              int32_t max_check, int search_dup,
+                              int search_deleted, int32_t* out_vids,
+                              float* out_dists)
+{
     SearchSpace sp;
     ndheap_init(&sp.ng, max_check * 30);
     ndheap_init(&sp.spt, max_check * 10);
@@ -487,22 +502,29 @@ int32_t orc_search(const OrcIndex* ix, const void* q, int32_t k,
             if (checkNode < -1) {
                 /* duplicate-center chain: the row's last slot points at the
                  * collapsed BKT node (-2 - treeIdx; NeighborhoodGraph.h:399,
-                 * BKTree.h:601-608); walk center + duplicates, AddPoint each
-                 * until it rejects (CheckDup = !AddPoint, BKTIndex.cpp:443). */
+                 * BKTree.h:601-608). searchDuplicated (CheckDup): walk
+                 * center + duplicates, AddPoint each until it rejects
+                 * (BKTIndex.cpp:443); NeverDup (refine path): AddPoint the
+                 * center only (StaticDispatch::NeverDup returns true). */
                 const BktNode* tnode = &ix->tree[-2 - checkNode];
                 int32_t i = -tnode->childStart;
                 do {
-                    if (not_deleted(ix, tmpNode)) {
-                        if (!qrs_add(&query, tmpNode, gnode.distance)) break;
+                    if ((search_deleted || not_deleted(ix, tmpNode))) {
+                        if (search_dup) {
+                            if (!qrs_add(&query, tmpNode, gnode.distance)) break;
+                        } else {
+                            qrs_add(&query, tmpNode, gnode.distance);
+                            break;
+                        }
                     }
                     if (i <= 0) break;
                     tmpNode = ix->tree[i].centerid;
                 } while (i++ < tnode->childEnd);
             } else {
-                if (not_deleted(ix, tmpNode)) qrs_add(&query, tmpNode, gnode.distance);
+                if ((search_deleted || not_deleted(ix, tmpNode))) qrs_add(&query, tmpNode, gnode.distance);
             }
         } else {
-            if (not_deleted(ix, tmpNode)) {
+            if ((search_deleted || not_deleted(ix, tmpNode))) {
                 if (gnode.distance > distpq_worst(&sp.results) || sp.checked > sp.max_check) {
                     finished = 1;
                     break;
@@ -822,6 +844,136 @@ void orc_truth(const OrcIndex* ix, const void* queries, int32_t nq, int32_t k,
 }
 
 /* ------------------------------------------------------------------ *
+ * Online add — exact restatement of BKT::Index<T>::AddIndex
+ * (src/Core/BKT/BKTIndex.cpp:902-970) for the sub-rebuild-threshold case
+ * (< AddCountForRebuild adds: the reference would only schedule its
+ * background tree rebuild beyond that), with
+ * NeighborhoodGraph::RefineNode (NeighborhoodGraph.h:535: RefineSearchIndex
+ * = search at MaxCheckForRefineGraph with searchDuplicated=false, k=CEF+1;
+ * then RebuildNeighbors) and RelativeNeighborhoodGraph::InsertNeighbors
+ * (RelativeNeighborhoodGraph.h:37-80) two-way edge updates.
+ * ------------------------------------------------------------------ */
+
+enum { ORC_REFINE_MAXCHECK = 8192 };  /* MaxCheckForRefineGraph default */
+
+/* RebuildNeighbors (RelativeNeighborhoodGraph.h:18-35) */
+static void orc_rebuild_neighbors(OrcIndex* ix, int32_t node,
+                                  const int32_t* res_vids, const float* res_dists,
+                                  int num_results)
+{
+    int32_t* nodes = ix->graph + (size_t)node * ix->deg;
+    int count = 0;
+    for (int j = 0; j < num_results && count < ix->deg; j++) {
+        if (res_vids[j] < 0) break;
+        if (res_vids[j] == node) continue;
+        int good = 1;
+        for (int kk = 0; kk < count; kk++) {
+            float d = orc_distance(ix->valuetype, ix->distmethod,
+                                   vec_at(ix, nodes[kk]), vec_at(ix, res_vids[j]),
+                                   ix->dim);
+            if (1.0f * d < res_dists[j]) { good = 0; break; }  /* RNGFactor=1 */
+        }
+        if (good) nodes[count++] = res_vids[j];
+    }
+    for (int j = count; j < ix->deg; j++) nodes[j] = -1;
+}
+
+/* InsertNeighbors (RelativeNeighborhoodGraph.h:37-80) */
+static void orc_insert_neighbors(OrcIndex* ix, int32_t node, int32_t insertNode,
+                                 float insertDist)
+{
+    int32_t* nodes = ix->graph + (size_t)node * ix->deg;
+    const void* nodeVec = vec_at(ix, node);
+    const void* insertVec = vec_at(ix, insertNode);
+    int checkSize = (nodes[ix->deg - 1] < -1) ? ix->deg - 1 : ix->deg;
+    for (int k = 0; k < checkSize; k++) {
+        int32_t tmpNode = nodes[k];
+        if (tmpNode < 0) { nodes[k] = insertNode; break; }
+        const void* tmpVec = vec_at(ix, tmpNode);
+        float tmpDist = orc_distance(ix->valuetype, ix->distmethod, tmpVec,
+                                     nodeVec, ix->dim);
+        if (tmpDist > insertDist ||
+            (insertDist == tmpDist && insertNode < tmpNode)) {
+            nodes[k] = insertNode;
+            while (++k < checkSize &&
+                   orc_distance(ix->valuetype, ix->distmethod, tmpVec, nodeVec,
+                                ix->dim) <=
+                   orc_distance(ix->valuetype, ix->distmethod, tmpVec, insertVec,
+                                ix->dim)) {
+                int32_t t = tmpNode; tmpNode = nodes[k]; nodes[k] = t;
+                if (tmpNode < 0) return;
+                tmpVec = vec_at(ix, tmpNode);
+            }
+            break;
+        } else if (orc_distance(ix->valuetype, ix->distmethod, tmpVec, insertVec,
+                                ix->dim) < insertDist) {
+            break;
+        }
+    }
+}
+
+int orc_add(OrcIndex* ix, const void* vectors, int32_t nadd, int32_t add_cef,
+            int normalized)
+{
+    if (ix->algo != 0 || nadd <= 0) return -1;
+    int32_t begin = ix->n, end = ix->n + nadd;
+    size_t esz = ix->esz;
+    ix->vectors = realloc(ix->vectors, (size_t)end * ix->dim * esz);
+    memcpy((char*)ix->vectors + (size_t)begin * ix->dim * esz, vectors,
+           (size_t)nadd * ix->dim * esz);
+    ix->graph = (int32_t*)realloc(ix->graph, (size_t)end * ix->deg * 4);
+    for (size_t i = (size_t)begin * ix->deg; i < (size_t)end * ix->deg; i++)
+        ix->graph[i] = -1;
+    if (ix->deleted) {
+        ix->deleted = (uint8_t*)realloc(ix->deleted, (size_t)end);
+        memset(ix->deleted + begin, 0, (size_t)nadd);
+    }
+    ix->n = end;
+
+    if (ix->distmethod == ORC_DM_COSINE && !normalized) {
+        /* Utils::Normalize (CommonUtils.h:62): scale to norm=base with
+         * C-cast truncation for integral types */
+        for (int32_t i = begin; i < end; i++) {
+            if (ix->valuetype == ORC_VT_FLOAT) {
+                float* v = (float*)vec_at(ix, i);
+                double s = 0;
+                for (int d = 0; d < ix->dim; d++) s += (double)v[d] * v[d];
+                s = sqrt(s);
+                if (s > 0)
+                    for (int d = 0; d < ix->dim; d++) v[d] = (float)(v[d] / s);
+            } else {
+                int8_t* v = (int8_t*)vec_at(ix, i);
+                double s = 0;
+                for (int d = 0; d < ix->dim; d++) s += (double)v[d] * v[d];
+                s = sqrt(s);
+                if (s > 0)
+                    for (int d = 0; d < ix->dim; d++)
+                        v[d] = (int8_t)(v[d] * 127.0 / s);
+            }
+        }
+    }
+
+    int k = add_cef + 1;
+    int32_t* rv = (int32_t*)malloc((size_t)k * 4);
+    float* rd = (float*)malloc((size_t)k * 4);
+    for (int32_t node = begin; node < end; node++) {
+        /* RefineNode(this, node, updateNeighbors=true, searchDeleted=true,
+         * AddCEF) — BKTIndex.cpp:968 */
+        orc_search_bkt(ix, vec_at(ix, node), k, ORC_REFINE_MAXCHECK,
+                       /*search_dup=*/0, /*search_deleted=*/1, rv, rd);
+        orc_rebuild_neighbors(ix, node, rv, rd, k);
+        for (int j = 0; j < k; j++) {
+            if (rv[j] < 0) break;
+            if (rv[j] == node) continue;
+            orc_insert_neighbors(ix, rv[j], node, rd[j]);
+        }
+    }
+    free(rv);
+    free(rd);
+    return 0;
+}
+
+/* ------------------------------------------------------------------ *
  * Construction / loading
  * ------------------------------------------------------------------ */
 
@@ -903,6 +1055,9 @@ void orc_free_index(OrcIndex* ix)
     free(ix->vectors); free(ix->tree_start); free(ix->tree); free(ix->kdt);
     free(ix->graph); free(ix->deleted); free(ix);
 }
+
+const int32_t* orc_graph_ptr(const OrcIndex* ix) { return ix->graph; }
+const void* orc_vectors_ptr(const OrcIndex* ix) { return ix->vectors; }
 
 int32_t orc_num_vectors(const OrcIndex* ix) { return ix->n; }
 int32_t orc_dim(const OrcIndex* ix) { return ix->dim; }

@@ -60,6 +60,9 @@ int32_t orc_dim(const OrcIndex* idx);
 int orc_valuetype(const OrcIndex* idx);
 int orc_distmethod(const OrcIndex* idx);
 int32_t orc_degree(const OrcIndex* idx);
+/* read-only views of the live index (tests: post-add graph comparison) */
+const int32_t* orc_graph_ptr(const OrcIndex* idx);
+const void* orc_vectors_ptr(const OrcIndex* idx);
 
 /* Exact restatement of BKT::Index<T>::SearchIndex for one query
  * (reference src/Core/BKT/BKTIndex.cpp:272-352 Search<> +
@@ -81,6 +84,12 @@ void orc_search_batch(const OrcIndex* idx, const void* queries, int32_t nq,
  * same distance function, top-k by (dist, vid)). */
 void orc_truth(const OrcIndex* idx, const void* queries, int32_t nq, int32_t k,
                int nthreads, int32_t* out_vids, float* out_dists);
+
+/* Online add — reference BKT AddIndex (BKTIndex.cpp:902-970), valid below
+ * the tree-rebuild threshold (AddCountForRebuild). add_cef: reference
+ * AddCEF default 500. */
+int orc_add(OrcIndex* idx, const void* vectors, int32_t nadd, int32_t add_cef,
+            int normalized);
 
 /* Iterative (streaming) search — reference SearchIterative /
  * ResultIterator (BKTIndex.cpp:354-427, ResultIterator.cpp). BKT only. */

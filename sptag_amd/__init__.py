@@ -78,6 +78,9 @@ def load_library():
         fn = getattr(lib, "sptag_amd_" + f)
         fn.restype = ctypes.c_int32
         fn.argtypes = [ctypes.c_void_p]
+    lib.sptag_amd_add.restype = ctypes.c_int
+    lib.sptag_amd_add.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
+                                  ctypes.c_int32, ctypes.c_int]
     lib.sptag_amd_delete.restype = ctypes.c_int
     lib.sptag_amd_delete.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
                                      ctypes.c_int32]
@@ -268,6 +271,19 @@ class AnnIndex:
         if rc != 0:
             raise SptagAmdError(rc, "truth")
         return vids, dists
+
+    def Add(self, vectors, normalized=False):
+        """Online add (reference AddIndex semantics below the tree-rebuild
+        threshold); runs GPU refine searches per added vector."""
+        vectors = np.ascontiguousarray(vectors,
+                                       dtype=_np_dtype(self.valuetype))
+        if vectors.ndim == 1:
+            vectors = vectors[None, :]
+        rc = self._lib.sptag_amd_add(
+            self._h, vectors.ctypes.data_as(ctypes.c_void_p),
+            vectors.shape[0], 1 if normalized else 0)
+        if rc != 0:
+            raise SptagAmdError(rc, "add")
 
     def Delete(self, vids):
         """Flag vector ids as deleted (reference DeleteIndex semantics);

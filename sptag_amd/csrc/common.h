@@ -13,7 +13,7 @@ enum { DM_L2 = 0, DM_COSINE = 1 };
 enum { DEFAULT_MAXCHECK = 8192, INIT_PIVOTS = 50, OTHER_PIVOTS = 4 };
 
 /* hard caps of the v1 kernel (host validates before launch) */
-enum { MAX_DEG = 64, MAX_K = 64, MAX_DIM = 4096 };
+enum { MAX_DEG = 64, MAX_K = 512, MAX_DIM = 4096 };
 
 enum { ALGO_BKT = 0, ALGO_KDT = 1 };
 
@@ -35,6 +35,8 @@ struct SearchCfg {
     int32_t nq, k, max_check;
     int32_t init_pivots, other_pivots;
     int32_t nobetter_threshold;  /* KDT: ThresholdOfNumberOfContinuousNoBetterPropagation */
+    int32_t search_dup;          /* searchDuplicated (BKT dispatch bit 1) */
+    int32_t search_deleted;      /* searchDeleted (dispatch bit 2) */
     int32_t ng_cap, spt_cap, dpq_cap;  /* heap capacities (entries) */
     int32_t vcap;                      /* visited table slots (pow2) */
 };

@@ -40,6 +40,7 @@ struct HostIndex {
     int32_t default_maxcheck = DEFAULT_MAXCHECK;
     int32_t init_pivots = INIT_PIVOTS, other_pivots = OTHER_PIVOTS;
     int32_t nobetter_threshold = 3;  /* KDT ParameterDefinitionList */
+    bool refine_mode = false;  /* internal: RefineSearchIndex dispatch flags */
     bool has_deleted = false;
     /* device allocations */
     void* d_vectors = nullptr;
@@ -394,6 +395,9 @@ static int search_device_core(SptagAmdIndex* ix, const void* d_q, int32_t nq,
     cfg.init_pivots = ix->init_pivots;
     cfg.other_pivots = ix->other_pivots;
     cfg.nobetter_threshold = ix->nobetter_threshold;
+    cfg.search_dup = 1;        /* SearchIndex(QueryResult&,bool) defaults */
+    cfg.search_deleted = 0;
+    if (ix->refine_mode) { cfg.search_dup = 0; cfg.search_deleted = 1; }
     cfg.dpq_cap = std::max(max_check / 16, k);   /* WorkSpace.h:268 */
     cfg.vcap = (int32_t)next_pow2((uint32_t)std::max(4096, max_check * 4));
     /* LDS-variant capacities: sized to the TYPICAL traversal occupancy so
@@ -706,6 +710,8 @@ int sptag_amd_iter_next(SptagAmdIterBatch* it, int32_t batch,
     cfg.init_pivots = ix->init_pivots;
     cfg.other_pivots = ix->other_pivots;
     cfg.nobetter_threshold = ix->nobetter_threshold;
+    cfg.search_dup = 1;
+    cfg.search_deleted = 0;
     cfg.dpq_cap = std::max(it->max_check / 16, batch);  /* ResetResult cap */
     cfg.vcap = it->vcap;
     cfg.ng_cap = it->ng_cap;
@@ -764,6 +770,227 @@ int sptag_amd_truth(SptagAmdIndex* ix, const void* queries, int32_t nq,
         rc = SPTAG_AMD_OK;
     }
     hipFree(d_q); hipFree(d_v); hipFree(d_d);
+    return rc;
+}
+
+/* Host-side distance restatements for the add path's edge arithmetic
+ * (RebuildNeighbors / InsertNeighbors recompute pair distances on the
+ * host, as the reference does on CPU): same rounding contracts as the
+ * kernels/oracle (fused 16-lane order for f32, exact ints for int8). */
+static float host_dist(const SptagAmdIndex* ix, const void* a, const void* b)
+{
+    int d = ix->dim;
+    if (ix->vt == VT_INT8) {
+        const int8_t* x = (const int8_t*)a;
+        const int8_t* y = (const int8_t*)b;
+        int32_t s = 0;
+        if (ix->dm == DM_L2) {
+            for (int i = 0; i < d; i++) { int32_t t = (int32_t)x[i] - y[i]; s += t * t; }
+            return (float)s;
+        }
+        for (int i = 0; i < d; i++) s += (int32_t)x[i] * y[i];
+        return (float)(16129 - s);
+    }
+    const float* x = (const float*)a;
+    const float* y = (const float*)b;
+    float a16[16], a8[8], a4[4];
+    int i = 0, j;
+    for (j = 0; j < 16; j++) a16[j] = 0.0f;
+    int nd16 = (d >> 4) << 4;
+    for (; i < nd16; i += 16)
+        for (j = 0; j < 16; j++) {
+            float xx = x[i + j], yy = y[i + j];
+            a16[j] = ix->dm == DM_L2 ? fmaf(xx - yy, xx - yy, a16[j])
+                                     : fmaf(xx, yy, a16[j]);
+        }
+    for (j = 0; j < 8; j++) a8[j] = a16[j] + a16[j + 8];
+    int nd8 = (d >> 3) << 3;
+    for (; i < nd8; i += 8)
+        for (j = 0; j < 8; j++) {
+            float xx = x[i + j], yy = y[i + j];
+            a8[j] = ix->dm == DM_L2 ? fmaf(xx - yy, xx - yy, a8[j])
+                                    : fmaf(xx, yy, a8[j]);
+        }
+    for (j = 0; j < 4; j++) a4[j] = a8[j] + a8[j + 4];
+    int nd4 = (d >> 2) << 2;
+    for (; i < nd4; i += 4)
+        for (j = 0; j < 4; j++) {
+            float xx = x[i + j], yy = y[i + j];
+            a4[j] = ix->dm == DM_L2 ? fmaf(xx - yy, xx - yy, a4[j])
+                                    : fmaf(xx, yy, a4[j]);
+        }
+    float diff = ((a4[0] + a4[1]) + a4[2]) + a4[3];
+    for (; i < d; i++) {
+        float xx = x[i], yy = y[i];
+        diff = ix->dm == DM_L2 ? fmaf(xx - yy, xx - yy, diff) : fmaf(xx, yy, diff);
+    }
+    return ix->dm == DM_L2 ? diff : 1.0f - diff;
+}
+
+static const void* hvec(const SptagAmdIndex* ix, int32_t v)
+{
+    return ix->h_vectors.data() + (size_t)v * ix->dim * ix->esz();
+}
+
+/* RelativeNeighborhoodGraph::RebuildNeighbors (RelativeNeighborhoodGraph.h:18) */
+static void host_rebuild_neighbors(SptagAmdIndex* ix, int32_t node,
+                                   const int32_t* rv, const float* rd, int nres)
+{
+    int32_t* nodes = ix->h_graph.data() + (size_t)node * ix->deg;
+    int count = 0;
+    for (int j = 0; j < nres && count < ix->deg; j++) {
+        if (rv[j] < 0) break;
+        if (rv[j] == node) continue;
+        bool good = true;
+        for (int kk = 0; kk < count; kk++) {
+            if (host_dist(ix, hvec(ix, nodes[kk]), hvec(ix, rv[j])) < rd[j]) {
+                good = false;
+                break;
+            }
+        }
+        if (good) nodes[count++] = rv[j];
+    }
+    for (int j = count; j < ix->deg; j++) nodes[j] = -1;
+}
+
+/* RelativeNeighborhoodGraph::InsertNeighbors (RelativeNeighborhoodGraph.h:37) */
+static void host_insert_neighbors(SptagAmdIndex* ix, int32_t node,
+                                  int32_t insertNode, float insertDist)
+{
+    int32_t* nodes = ix->h_graph.data() + (size_t)node * ix->deg;
+    const void* nodeVec = hvec(ix, node);
+    const void* insertVec = hvec(ix, insertNode);
+    int checkSize = (nodes[ix->deg - 1] < -1) ? ix->deg - 1 : ix->deg;
+    for (int k = 0; k < checkSize; k++) {
+        int32_t tmpNode = nodes[k];
+        if (tmpNode < 0) { nodes[k] = insertNode; break; }
+        const void* tmpVec = hvec(ix, tmpNode);
+        float tmpDist = host_dist(ix, tmpVec, nodeVec);
+        if (tmpDist > insertDist ||
+            (insertDist == tmpDist && insertNode < tmpNode)) {
+            nodes[k] = insertNode;
+            while (++k < checkSize &&
+                   host_dist(ix, tmpVec, nodeVec) <= host_dist(ix, tmpVec, insertVec)) {
+                std::swap(tmpNode, nodes[k]);
+                if (tmpNode < 0) return;
+                tmpVec = hvec(ix, tmpNode);
+            }
+            break;
+        } else if (host_dist(ix, tmpVec, insertVec) < insertDist) {
+            break;
+        }
+    }
+}
+
+int sptag_amd_add(SptagAmdIndex* ix, const void* vectors, int32_t nadd,
+                  int normalized)
+{
+    if (!ix || !vectors || nadd <= 0) return SPTAG_AMD_ERR_PARAM;
+    if (ix->algo != ALGO_BKT) return SPTAG_AMD_ERR_UNSUPP;
+    if (!sptag_amd_gpu_available() || !ix->d_vectors) {
+        fprintf(stderr, "sptag_amd: add requires a HIP device (refine searches)\n");
+        return SPTAG_AMD_ERR_NOGPU;
+    }
+    std::lock_guard<std::mutex> g(ix->lock);
+    HIP_OR_FAIL(hipSetDevice(ix->device), SPTAG_AMD_ERR_NOGPU);
+
+    int32_t begin = ix->n, end = ix->n + nadd;
+    size_t esz = ix->esz();
+    /* host copies (AddIndex: m_pSamples/m_pGraph/m_deletedID AddBatch) */
+    ix->h_vectors.resize((size_t)end * ix->dim * esz);
+    memcpy(ix->h_vectors.data() + (size_t)begin * ix->dim * esz, vectors,
+           (size_t)nadd * ix->dim * esz);
+    ix->h_graph.resize((size_t)end * ix->deg, -1);
+    if (!ix->h_deleted.empty()) ix->h_deleted.resize((size_t)end, 0);
+    ix->n = end;
+
+    if (ix->dm == DM_COSINE && !normalized) {
+        /* Utils::Normalize to norm=base, C-cast truncation for int8
+         * (CommonUtils.h:62) */
+        for (int32_t i = begin; i < end; i++) {
+            char* vp = ix->h_vectors.data() + (size_t)i * ix->dim * esz;
+            double s = 0;
+            if (ix->vt == VT_FLOAT) {
+                float* v = (float*)vp;
+                for (int d = 0; d < ix->dim; d++) s += (double)v[d] * v[d];
+                s = sqrt(s);
+                if (s > 0) for (int d = 0; d < ix->dim; d++) v[d] = (float)(v[d] / s);
+            } else {
+                int8_t* v = (int8_t*)vp;
+                for (int d = 0; d < ix->dim; d++) s += (double)v[d] * v[d];
+                s = sqrt(s);
+                if (s > 0) for (int d = 0; d < ix->dim; d++)
+                    v[d] = (int8_t)(v[d] * 127.0 / s);
+            }
+        }
+    }
+
+    /* device: re-allocate vectors+graph at the new size */
+    void* nv = nullptr;
+    int32_t* ng = nullptr;
+    HIP_OR_FAIL(hipMalloc(&nv, ix->h_vectors.size()), SPTAG_AMD_ERR_OOM);
+    HIP_OR_FAIL(hipMemcpy(nv, ix->h_vectors.data(), ix->h_vectors.size(),
+                          hipMemcpyHostToDevice), SPTAG_AMD_ERR_NOGPU);
+    HIP_OR_FAIL(hipMalloc(&ng, ix->h_graph.size() * 4), SPTAG_AMD_ERR_OOM);
+    HIP_OR_FAIL(hipMemcpy(ng, ix->h_graph.data(), ix->h_graph.size() * 4,
+                          hipMemcpyHostToDevice), SPTAG_AMD_ERR_NOGPU);
+    (void)hipFree(ix->d_vectors);
+    (void)hipFree(ix->d_graph);
+    ix->d_vectors = nv;
+    ix->d_graph = ng;
+    if (ix->has_deleted) {
+        (void)hipFree(ix->d_deleted);
+        HIP_OR_FAIL(hipMalloc(&ix->d_deleted, ix->h_deleted.size()), SPTAG_AMD_ERR_OOM);
+        HIP_OR_FAIL(hipMemcpy(ix->d_deleted, ix->h_deleted.data(),
+                              ix->h_deleted.size(), hipMemcpyHostToDevice),
+                    SPTAG_AMD_ERR_NOGPU);
+    }
+
+    /* per new node: RefineNode (search CEF+1 at MaxCheckForRefineGraph with
+     * the refine dispatch flags, then RNG rebuild + two-way inserts).
+     * Sequential, as the reference's add loop is (BKTIndex.cpp:966-969);
+     * each refine sees the edges of previously added nodes. */
+    const int k = 500 + 1;   /* AddCEF default (ParameterDefinitionList) */
+    std::vector<int32_t> rv(k);
+    std::vector<float> rd(k);
+    int32_t* d_v = nullptr;
+    float* d_d = nullptr;
+    void* d_q = nullptr;
+    HIP_OR_FAIL(hipMalloc(&d_v, (size_t)k * 4), SPTAG_AMD_ERR_OOM);
+    HIP_OR_FAIL(hipMalloc(&d_d, (size_t)k * 4), SPTAG_AMD_ERR_OOM);
+    HIP_OR_FAIL(hipMalloc(&d_q, (size_t)ix->dim * esz), SPTAG_AMD_ERR_OOM);
+    int rc = SPTAG_AMD_OK;
+    for (int32_t node = begin; node < end && rc == SPTAG_AMD_OK; node++) {
+        HIP_OR_FAIL(hipMemcpy(d_q, hvec(ix, node), (size_t)ix->dim * esz,
+                              hipMemcpyHostToDevice), SPTAG_AMD_ERR_NOGPU);
+        ix->refine_mode = true;
+        rc = search_device_core(ix, d_q, 1, k, 8192 /*MaxCheckForRefineGraph*/,
+                                d_v, d_d);
+        ix->refine_mode = false;
+        if (rc != SPTAG_AMD_OK) break;
+        HIP_OR_FAIL(hipMemcpy(rv.data(), d_v, (size_t)k * 4, hipMemcpyDeviceToHost),
+                    SPTAG_AMD_ERR_NOGPU);
+        HIP_OR_FAIL(hipMemcpy(rd.data(), d_d, (size_t)k * 4, hipMemcpyDeviceToHost),
+                    SPTAG_AMD_ERR_NOGPU);
+        host_rebuild_neighbors(ix, node, rv.data(), rd.data(), k);
+        /* upload the refined row, then the two-way inserts */
+        HIP_OR_FAIL(hipMemcpy(ix->d_graph + (size_t)node * ix->deg,
+                              ix->h_graph.data() + (size_t)node * ix->deg,
+                              (size_t)ix->deg * 4, hipMemcpyHostToDevice),
+                    SPTAG_AMD_ERR_NOGPU);
+        for (int j = 0; j < k; j++) {
+            if (rv[j] < 0) break;
+            if (rv[j] == node) continue;
+            host_insert_neighbors(ix, rv[j], node, rd[j]);
+            HIP_OR_FAIL(hipMemcpy(ix->d_graph + (size_t)rv[j] * ix->deg,
+                                  ix->h_graph.data() + (size_t)rv[j] * ix->deg,
+                                  (size_t)ix->deg * 4, hipMemcpyHostToDevice),
+                        SPTAG_AMD_ERR_NOGPU);
+        }
+    }
+    (void)hipFree(d_v);
+    (void)hipFree(d_d);
+    (void)hipFree(d_q);
     return rc;
 }
 

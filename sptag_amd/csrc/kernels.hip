@@ -602,26 +602,34 @@ void bkt_search_kernel(DevIndex di, SearchCfg cfg, SearchBufs bufs)
         int32_t checkNode = __shfl(nn, checkPos);
 
         if (lane == 0) {
-            /* BKTIndex.cpp:290-331: result/termination block */
+            /* BKTIndex.cpp:290-331: result/termination block. The dispatch
+             * flags (BKTIndex.cpp:471-507): searchDeleted makes the delete
+             * filter AlwaysTrue; searchDuplicated selects CheckDup (walk
+             * the duplicate chain until AddPoint rejects) vs NeverDup (add
+             * the center only). */
             if (gnode.distance <= qrs[0].dist) {
                 if (checkNode < -1) {
-                    /* duplicate-center chain (BKTIndex.cpp:292-312) */
                     const int32_t* tn = &di.tree_nodes[(size_t)(-2 - checkNode) * 3];
                     int32_t i = -tn[1];
                     int32_t tmpNode = gnode.node;
                     do {
-                        if (not_deleted(di, tmpNode)) {
-                            if (!qrs_add(qrs, cfg.k, tmpNode, gnode.distance)) break;
+                        if (cfg.search_deleted || not_deleted(di, tmpNode)) {
+                            if (cfg.search_dup) {
+                                if (!qrs_add(qrs, cfg.k, tmpNode, gnode.distance)) break;
+                            } else {
+                                qrs_add(qrs, cfg.k, tmpNode, gnode.distance);
+                                break;
+                            }
                         }
                         if (i <= 0) break;
                         tmpNode = di.tree_nodes[(size_t)i * 3];
                     } while (i++ < tn[2]);
                 } else {
-                    if (not_deleted(di, gnode.node))
+                    if (cfg.search_deleted || not_deleted(di, gnode.node))
                         qrs_add(qrs, cfg.k, gnode.node, gnode.distance);
                 }
             } else {
-                if (not_deleted(di, gnode.node)) {
+                if (cfg.search_deleted || not_deleted(di, gnode.node)) {
                     if (gnode.distance > dpq[1] || ss->checked > cfg.max_check)
                         ss->terminate = 1;
                 }
@@ -681,9 +689,9 @@ void bkt_search_kernel(DevIndex di, SearchCfg cfg, SearchBufs bufs)
         qrs_sort(qrs, cfg.k);
     }
     __syncthreads();
-    if (lane < cfg.k) {
-        bufs.out_vids[(size_t)q * cfg.k + lane] = qrs[lane].vid;
-        bufs.out_dists[(size_t)q * cfg.k + lane] = qrs[lane].dist;
+    for (int i = lane; i < cfg.k; i += 64) {
+        bufs.out_vids[(size_t)q * cfg.k + i] = qrs[i].vid;
+        bufs.out_dists[(size_t)q * cfg.k + i] = qrs[i].dist;
     }
 }
 
@@ -848,9 +856,9 @@ void bkt_iter_kernel(DevIndex di, SearchCfg cfg, IterBufs ib, int batch)
         ib.out_relaxed[q] = st.relaxed;
     }
     __syncthreads();
-    if (lane < batch) {
-        ib.out_vids[(size_t)q * batch + lane] = qrs[lane].vid;
-        ib.out_dists[(size_t)q * batch + lane] = qrs[lane].dist;
+    for (int i = lane; i < batch; i += 64) {
+        ib.out_vids[(size_t)q * batch + i] = qrs[i].vid;
+        ib.out_dists[(size_t)q * batch + i] = qrs[i].dist;
     }
 }
 
@@ -1050,9 +1058,9 @@ void kdt_search_kernel(DevIndex di, SearchCfg cfg, SearchBufs bufs)
         qrs_sort(qrs, cfg.k);
     }
     __syncthreads();
-    if (lane < cfg.k) {
-        bufs.out_vids[(size_t)q * cfg.k + lane] = qrs[lane].vid;
-        bufs.out_dists[(size_t)q * cfg.k + lane] = qrs[lane].dist;
+    for (int i = lane; i < cfg.k; i += 64) {
+        bufs.out_vids[(size_t)q * cfg.k + i] = qrs[i].vid;
+        bufs.out_dists[(size_t)q * cfg.k + i] = qrs[i].dist;
     }
 }
 
@@ -1102,9 +1110,9 @@ void truth_kernel(DevIndex di, const void* queries, int32_t nq, int32_t k,
     }
     if (lane == 0) qrs_sort(qrs, k);
     __syncthreads();
-    if (lane < k) {
-        out_vids[(size_t)q * k + lane] = qrs[lane].vid;
-        out_dists[(size_t)q * k + lane] = qrs[lane].dist;
+    for (int i = lane; i < k; i += 64) {
+        out_vids[(size_t)q * k + i] = qrs[i].vid;
+        out_dists[(size_t)q * k + i] = qrs[i].dist;
     }
 }
 
