@@ -144,6 +144,10 @@ int sptag_amd_add(SptagAmdIndex* idx, const void* vectors, int32_t nadd,
  * filter them with the reference's CheckIfNotDeleted dispatch (already
  * covered by the parity suite). Persisted by save_index in deletes.bin. */
 int sptag_amd_delete(SptagAmdIndex* idx, const int32_t* vids, int32_t n);
+/* the by-vector overload (BKTIndex.cpp:876: search CEF results, delete the
+ * exact duplicates at distance < 1e-6) */
+int sptag_amd_delete_by_vector(SptagAmdIndex* idx, const void* vectors,
+                               int32_t n);
 int64_t sptag_amd_deleted_count(const SptagAmdIndex* idx);
 
 /* Write the index back out in the reference's byte format

@@ -84,6 +84,8 @@ def load_library():
     lib.sptag_amd_delete.restype = ctypes.c_int
     lib.sptag_amd_delete.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
                                      ctypes.c_int32]
+    lib.sptag_amd_delete_by_vector.restype = ctypes.c_int
+    lib.sptag_amd_delete_by_vector.argtypes = lib.sptag_amd_delete.argtypes
     lib.sptag_amd_deleted_count.restype = ctypes.c_int64
     lib.sptag_amd_deleted_count.argtypes = [ctypes.c_void_p]
     lib.sptag_amd_iter_create.restype = ctypes.c_void_p
@@ -293,6 +295,18 @@ class AnnIndex:
             self._h, vids.ctypes.data_as(ctypes.c_void_p), vids.size)
         if rc != 0:
             raise SptagAmdError(rc, "delete")
+
+    def DeleteByVector(self, vectors):
+        """Delete exact duplicates of the given vectors (reference
+        DeleteIndex(const void*, n) semantics: CEF search + dist < 1e-6)."""
+        vectors = np.ascontiguousarray(vectors,
+                                       dtype=_np_dtype(self.valuetype))
+        if vectors.ndim == 1:
+            vectors = vectors[None, :]
+        rc = self._lib.sptag_amd_delete_by_vector(
+            self._h, vectors.ctypes.data_as(ctypes.c_void_p), vectors.shape[0])
+        if rc != 0:
+            raise SptagAmdError(rc, "delete_by_vector")
 
     @property
     def deleted_count(self):

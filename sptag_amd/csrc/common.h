@@ -13,7 +13,7 @@ enum { DM_L2 = 0, DM_COSINE = 1 };
 enum { DEFAULT_MAXCHECK = 8192, INIT_PIVOTS = 50, OTHER_PIVOTS = 4 };
 
 /* hard caps of the v1 kernel (host validates before launch) */
-enum { MAX_DEG = 64, MAX_K = 512, MAX_DIM = 4096 };
+enum { MAX_DEG = 64, MAX_K = 1024, MAX_DIM = 4096 };
 
 enum { ALGO_BKT = 0, ALGO_KDT = 1 };
 

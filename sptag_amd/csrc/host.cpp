@@ -994,6 +994,30 @@ int sptag_amd_add(SptagAmdIndex* ix, const void* vectors, int32_t nadd,
     return rc;
 }
 
+int sptag_amd_delete_by_vector(SptagAmdIndex* ix, const void* vectors,
+                               int32_t n)
+{
+    /* reference DeleteIndex(const void*, SizeType) (BKTIndex.cpp:876-891):
+     * search CEF=1000 results per vector, delete every id at distance
+     * < 1e-6 (exact duplicates of the given vector). */
+    if (!ix || !vectors || n <= 0) return SPTAG_AMD_ERR_PARAM;
+    if (ix->algo != ALGO_BKT) return SPTAG_AMD_ERR_UNSUPP;
+    if (!sptag_amd_gpu_available() || !ix->d_vectors) return SPTAG_AMD_ERR_NOGPU;
+    const int k = 1000;   /* m_iCEF default */
+    std::vector<int32_t> rv((size_t)n * k);
+    std::vector<float> rd((size_t)n * k);
+    int rc = sptag_amd_search_batch(ix, vectors, n, k, 0, rv.data(), rd.data());
+    if (rc != SPTAG_AMD_OK) return rc;
+    std::vector<int32_t> dels;
+    for (int32_t i = 0; i < n; i++)
+        for (int j = 0; j < k; j++) {
+            if (rv[(size_t)i * k + j] < 0) break;
+            if (rd[(size_t)i * k + j] < 1e-6f) dels.push_back(rv[(size_t)i * k + j]);
+        }
+    if (dels.empty()) return SPTAG_AMD_OK;
+    return sptag_amd_delete(ix, dels.data(), (int32_t)dels.size());
+}
+
 int sptag_amd_delete(SptagAmdIndex* ix, const int32_t* vids, int32_t n)
 {
     if (!ix || !vids || n <= 0) return SPTAG_AMD_ERR_PARAM;
