@@ -15,7 +15,8 @@ import os
 
 import numpy as np
 
-__all__ = ["AnnIndex", "load_library", "gpu_available", "SptagAmdError"]
+__all__ = ["AnnIndex", "MetadataSet", "load_library", "gpu_available",
+           "SptagAmdError"]
 
 _LIB = None
 _LIBPATH = os.environ.get(
@@ -132,7 +133,9 @@ class AnnIndex:
     def Load(cls, folder, device=0):
         lib = load_library()
         h = lib.sptag_amd_load_index(str(folder).encode(), device)
-        return cls(h)
+        ix = cls(h)
+        ix.metadata = MetadataSet.from_index_folder(str(folder))
+        return ix
 
     @classmethod
     def FromArrays(cls, vectors, tree_start, tree_nodes, graph, distmethod,
@@ -258,6 +261,15 @@ class AnnIndex:
         vids, dists = self.BatchSearch(query, k, max_check)
         return vids[0], dists[0]
 
+    def BatchSearchWithMeta(self, queries, k, max_check=0):
+        """reference AnnIndex.BatchSearchWithMetaData: results plus the
+        per-result metadata blobs (empty when the index has none)."""
+        vids, dists = self.BatchSearch(queries, k, max_check)
+        meta = getattr(self, "metadata", None)
+        blobs = [[meta.get(int(v)) if meta else b"" for v in row]
+                 for row in vids]
+        return vids, dists, blobs
+
     def Truth(self, queries, k):
         """Exact brute-force top-k on the GPU."""
         queries = np.ascontiguousarray(queries, dtype=_np_dtype(self.valuetype))
@@ -323,6 +335,49 @@ class AnnIndex:
         rc = self._lib.sptag_amd_save_index(self._h, str(folder).encode())
         if rc != 0:
             raise SptagAmdError(rc, "save_index")
+
+
+class MetadataSet:
+    """Reference MemMetadataSet file pair (src/Core/MetadataSet.cpp:269-283:
+    metadataIndex = [int32 count][uint64 offsets x (count+1)]; metadata =
+    concatenated byte blobs). Post-search VID -> bytes lookup — host-side
+    only, exactly as the reference's SearchIndex metadata copy
+    (BKTIndex.cpp:611-618)."""
+
+    def __init__(self, meta_file, metaindex_file):
+        with open(metaindex_file, "rb") as f:
+            self.count = int(np.frombuffer(f.read(4), dtype=np.int32)[0])
+            self.offsets = np.frombuffer(f.read(8 * (self.count + 1)),
+                                         dtype=np.uint64)
+        self.blob = open(meta_file, "rb").read()
+
+    def get(self, vid):
+        if vid < 0 or vid >= self.count:
+            return b""
+        return self.blob[int(self.offsets[vid]):int(self.offsets[vid + 1])]
+
+    @classmethod
+    def from_index_folder(cls, folder):
+        """Load the pair named by indexloader.ini's [MetaData] section
+        (VectorIndex.cpp:618 LoadIndex), or None when the index has none."""
+        ini = os.path.join(folder, "indexloader.ini")
+        if not os.path.exists(ini):
+            return None
+        meta = metaidx = None
+        section = None
+        for line in open(ini):
+            line = line.strip()
+            if line.startswith("["):
+                section = line
+            elif section == "[MetaData]" and "=" in line:
+                key, val = line.split("=", 1)
+                if key == "MetaDataFilePath":
+                    meta = val
+                elif key == "MetaDataIndexPath":
+                    metaidx = val
+        if not meta or not metaidx:
+            return None
+        return cls(os.path.join(folder, meta), os.path.join(folder, metaidx))
 
 
 class IterBatch:

@@ -146,3 +146,35 @@ def test_error_paths(tmp_path):
     d.mkdir()
     (d / "indexloader.ini").write_text("[Index]\nIndexAlgoType=SPANN\n")
     assert lib.sptag_amd_load_index(str(d).encode(), 0) is None
+
+
+def test_metadata_set(tmp_path):
+    """MemMetadataSet file-pair reader (reference MetadataSet.cpp:269-283)."""
+    from sptag_amd import MetadataSet
+    blobs = [b"alpha", b"", b"gamma-123"]
+    offs = np.zeros(len(blobs) + 1, dtype=np.uint64)
+    for i, b in enumerate(blobs):
+        offs[i + 1] = offs[i] + len(b)
+    (tmp_path / "metadata.bin").write_bytes(b"".join(blobs))
+    with open(tmp_path / "metadataIndex.bin", "wb") as f:
+        f.write(np.int32(len(blobs)).tobytes())
+        f.write(offs.tobytes())
+    ms = MetadataSet(str(tmp_path / "metadata.bin"),
+                     str(tmp_path / "metadataIndex.bin"))
+    assert [ms.get(i) for i in range(3)] == blobs
+    assert ms.get(-1) == b"" and ms.get(3) == b""
+
+    # an index folder whose ini names the pair
+    import shutil
+    g = load_golden("f32_l2_n10k_d32")
+    idx2 = tmp_path / "idx"
+    shutil.copytree(g["index"], idx2)
+    with open(idx2 / "indexloader.ini", "a") as f:
+        f.write("\n[MetaData]\nMetaDataFilePath=metadata.bin\n"
+                "MetaDataIndexPath=metadataIndex.bin\n")
+    shutil.copy(tmp_path / "metadata.bin", idx2 / "metadata.bin")
+    shutil.copy(tmp_path / "metadataIndex.bin", idx2 / "metadataIndex.bin")
+    from sptag_amd import AnnIndex
+    ix = AnnIndex.Load(str(idx2))
+    assert ix.metadata is not None
+    assert ix.metadata.get(2) == b"gamma-123"
