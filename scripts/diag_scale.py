@@ -49,7 +49,8 @@ def main():
         print(f"GPU    mc={mc}: recall={bench.recall_at_k(gv, tv, 10):.4f}",
               flush=True)
 
-    return  # oracle leg not needed once kernel==oracle is established
+    if os.environ.get("DIAG_ORACLE") != "1":
+        return   # kernel==oracle was established (agreement 1.000 at 30M)
     oix = OrcIndex.from_arrays(arrays["vectors"], arrays["tree_start"],
                                arrays["tree_nodes"], arrays["graph"], "Cosine")
     for mc in [4096, 16384]:
