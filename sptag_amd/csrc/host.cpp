@@ -761,15 +761,20 @@ int sptag_amd_truth(SptagAmdIndex* ix, const void* queries, int32_t nq,
     HIP_OR_FAIL(hipMalloc(&d_q, qbytes), SPTAG_AMD_ERR_OOM);
     HIP_OR_FAIL(hipMalloc(&d_v, (size_t)nq * k * 4), SPTAG_AMD_ERR_OOM);
     HIP_OR_FAIL(hipMalloc(&d_d, (size_t)nq * k * 4), SPTAG_AMD_ERR_OOM);
-    hipMemcpy(d_q, queries, qbytes, hipMemcpyHostToDevice);
-    int err = launch_truth(ix->vt, ix->dm, ix->dev(), d_q, nq, k, d_v, d_d, nullptr);
     int rc = SPTAG_AMD_ERR_NOGPU;
-    if (err == 0 && hipDeviceSynchronize() == hipSuccess) {
-        hipMemcpy(out_vids, d_v, (size_t)nq * k * 4, hipMemcpyDeviceToHost);
-        hipMemcpy(out_dists, d_d, (size_t)nq * k * 4, hipMemcpyDeviceToHost);
-        rc = SPTAG_AMD_OK;
+    if (hipMemcpy(d_q, queries, qbytes, hipMemcpyHostToDevice) == hipSuccess) {
+        int err = launch_truth(ix->vt, ix->dm, ix->dev(), d_q, nq, k, d_v, d_d,
+                               nullptr);
+        if (err == 0 && hipDeviceSynchronize() == hipSuccess &&
+            hipMemcpy(out_vids, d_v, (size_t)nq * k * 4,
+                      hipMemcpyDeviceToHost) == hipSuccess &&
+            hipMemcpy(out_dists, d_d, (size_t)nq * k * 4,
+                      hipMemcpyDeviceToHost) == hipSuccess)
+            rc = SPTAG_AMD_OK;
     }
-    hipFree(d_q); hipFree(d_v); hipFree(d_d);
+    (void)hipFree(d_q);
+    (void)hipFree(d_v);
+    (void)hipFree(d_d);
     return rc;
 }
 
