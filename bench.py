@@ -58,7 +58,7 @@ CONFIGS = {
     "bkt_100m_d100_i8_cos": dict(n=100_000_000, d=100, dtype="i8",
                                  metric="Cosine", nq=10_000, k=10,
                                  ncenters=16384, sigma=30.0, cand=128,
-                                 ntrees=6, refine=1),
+                                 ntrees=4, refine=0, srefine=1),
     # BASELINE.json configs[3] — KDT, embedding shape. Note: the KDT
     # algorithm's no-better-propagation termination caps recall on this
     # data family at ~0.93-0.94 for the REFERENCE implementation as well
@@ -298,7 +298,9 @@ def main():
         x_np, cfg["metric"], algo=cfg.get("algo", "BKT"),
         cand=args.cand or cfg.get("cand", 256), kdt_trees=2,
         ntrees=cfg.get("ntrees", args.ntrees),
-        refine_rounds=cfg.get("refine", args.refine), device=device, normalized=False,
+        refine_rounds=cfg.get("refine", args.refine),
+        search_refine_rounds=cfg.get("srefine", 0),
+        device=device, normalized=False,
         verbose=(rank == 0))
     log(rank, f"index built ({time.time()-t0:.1f}s)")
     torch.cuda.empty_cache()   # release build-phase cache so the extension's

@@ -24,12 +24,14 @@ def main():
     ntrees = int(sys.argv[2]) if len(sys.argv) > 2 else 2
     refine = int(sys.argv[3]) if len(sys.argv) > 3 else 0
     cand = int(sys.argv[4]) if len(sys.argv) > 4 else 96
+    srefine = int(sys.argv[5]) if len(sys.argv) > 5 else 0
     cfg = dict(bench.CONFIGS["bkt_100m_d100_i8_cos"])
     cfg["n"] = n
     x, q, lo = bench.gen_data(cfg, 0, 1, "cuda:0", torch)
     t0 = time.time()
     arrays = build_index_arrays(x.cpu().numpy(), "Cosine", cand=cand,
                                 ntrees=ntrees, refine_rounds=refine,
+                                search_refine_rounds=srefine,
                                 device="cuda:0")
     print(f"built {n} in {time.time()-t0:.0f}s", flush=True)
     torch.cuda.empty_cache()

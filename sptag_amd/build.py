@@ -672,10 +672,65 @@ def _exact_l2(xf, q_rows, c_ids):
     return d.masked_fill(c_ids < 0, float("inf"))
 
 
+def refine_via_search(vectors_t, tree_start, tree_nodes, graph, cand_ids,
+                      cand_dst, distmethod, *, algo="BKT", degree=32,
+                      cand=256, rounds=1, k=64, max_check=2048,
+                      rng_factor=1.0, chunk=1_000_000, verbose=False):
+    """The reference's own refinement recipe (NeighborhoodGraph.h:460-560
+    RefineGraph: every node re-searches the CURRENT index and its edges are
+    RNG-rebuilt from the results) — run on the PRODUCT GPU searcher via the
+    C-ABI. This is what creates the cross-cluster skip edges that truncated
+    candidate pools miss at 100M scale. GPU-only (the
+    searcher has no CPU path); vectors_t is the normalized torch tensor on
+    the device; cand lists are the builder's pools (merged in)."""
+    import sptag_amd
+    device = vectors_t.device
+    n = vectors_t.shape[0]
+    xf = vectors_t if vectors_t.dtype == torch.float32 else vectors_t.float()
+    self_ids = torch.arange(n, device=device, dtype=torch.int32)
+    x_np = vectors_t.cpu().numpy()
+    graph_t = torch.as_tensor(graph, device=device)         if not torch.is_tensor(graph) else graph
+    # search distances arrive in the index metric; the builder's pools are
+    # L2 on the (normalized) float view: for cosine, L2^2 = 2*d_cos exactly
+    # on base-normalized pairs, so a pure scale aligns them.
+    scale = 2.0 if distmethod == "Cosine" else 1.0
+    for r in range(rounds):
+        if algo == "KDT":
+            ix = sptag_amd.AnnIndex.FromArraysKDT(
+                x_np, tree_start, tree_nodes, graph_t.cpu().numpy(), distmethod)
+        else:
+            ix = sptag_amd.AnnIndex.FromArrays(
+                x_np, tree_start, tree_nodes, graph_t.cpu().numpy(), distmethod)
+        d_vids = torch.empty((chunk, k), dtype=torch.int32, device=device)
+        d_dists = torch.empty((chunk, k), dtype=torch.float32, device=device)
+        for s0 in range(0, n, chunk):
+            e0 = min(n, s0 + chunk)
+            B = e0 - s0
+            q = vectors_t[s0:e0].contiguous()
+            ix.BatchSearchDevice(q.data_ptr(), B, k, d_vids.data_ptr(),
+                                 d_dists.data_ptr(), max_check)
+            sv = d_vids[:B]
+            sd = d_dists[:B] * scale
+            sd = sd.masked_fill(sv < 0, float("inf"))
+            cand_ids[s0:e0], cand_dst[s0:e0] = _merge_candidates(
+                cand_ids[s0:e0], cand_dst[s0:e0], sv, sd, cand,
+                self_ids[s0:e0])
+            graph_t[s0:e0] = _rng_prune(xf, cand_ids[s0:e0], cand_dst[s0:e0],
+                                        degree, rng_factor, device)
+            if verbose and (s0 // chunk) % 10 == 0:
+                print(f"  search-refine round {r + 1}: {e0}/{n}")
+        del ix, d_vids, d_dists
+        torch.cuda.empty_cache()
+        if verbose:
+            print(f"  search-refine round {r + 1}/{rounds} done")
+    return graph_t, cand_ids, cand_dst
+
+
 def build_index_arrays(vectors, distmethod, *, algo="BKT", degree=32, ntrees=4,
                        tpt_leaf=1000, cand=256, kmeans_k=32, leaf_size=32,
-                       refine_rounds=0, kdt_trees=1, seed=2016, device=None,
-                       normalized=False, verbose=False):
+                       refine_rounds=0, search_refine_rounds=0, kdt_trees=1,
+                       seed=2016, device=None, normalized=False,
+                       verbose=False):
     """Full build: returns dict(vectors, tree_start, tree_nodes, graph) ready
     for AnnIndex.FromArrays (vectors already cosine-normalized when needed,
     as the reference stores them on disk)."""
@@ -694,11 +749,18 @@ def build_index_arrays(vectors, distmethod, *, algo="BKT", degree=32, ntrees=4,
         vectors, degree=degree, ntrees=ntrees, tpt_leaf=tpt_leaf, cand=cand,
         seed=seed, device=device, verbose=verbose)
     if refine_rounds > 0:
-        graph, _, _ = refine_graph(vectors, graph, cids, cdst, degree=degree,
-                                   cand=cand, rounds=refine_rounds,
-                                   device=device, seed=seed, verbose=verbose)
-    else:
-        graph = graph.cpu().numpy()
+        graph, cids, cdst = refine_graph(
+            vectors, graph, cids, cdst, degree=degree, cand=cand,
+            rounds=refine_rounds, device=device, seed=seed, verbose=verbose)
+        graph = torch.as_tensor(graph, device=device or _dev())
+    if search_refine_rounds > 0 and torch.cuda.is_available():
+        xt = torch.as_tensor(vectors, device=device or _dev())
+        graph, cids, cdst = refine_via_search(
+            xt, tree_start, tree_nodes, graph, cids, cdst, distmethod,
+            algo=algo, degree=degree, cand=cand,
+            rounds=search_refine_rounds, verbose=verbose)
+        del xt
+    graph = graph.cpu().numpy() if torch.is_tensor(graph) else graph
     return {"vectors": vectors, "tree_start": tree_start,
             "tree_nodes": tree_nodes, "graph": graph,
             "distmethod": distmethod, "algo": algo}
