@@ -674,13 +674,16 @@ def _exact_l2(xf, q_rows, c_ids):
 
 def refine_via_search(vectors_t, tree_start, tree_nodes, graph, cand_ids,
                       cand_dst, distmethod, *, algo="BKT", degree=32,
-                      cand=256, rounds=1, k=64, max_check=2048,
+                      cand=256, rounds=1, k=512, max_check=8192,
                       rng_factor=1.0, chunk=1_000_000, verbose=False):
     """The reference's own refinement recipe (NeighborhoodGraph.h:460-560
     RefineGraph: every node re-searches the CURRENT index and its edges are
     RNG-rebuilt from the results) — run on the PRODUCT GPU searcher via the
     C-ABI. This is what creates the cross-cluster skip edges that truncated
-    candidate pools miss at 100M scale. GPU-only (the
+    candidate pools miss at 100M scale. Depth matters: the
+    skip edges come from the DEEP tail of the result list (the reference
+    uses CEF=1000; k=64 measured ineffective at 30M — 0.664 vs 0.649
+    baseline), hence k=512 / max_check=8192 defaults. GPU-only (the
     searcher has no CPU path); vectors_t is the normalized torch tensor on
     the device; cand lists are the builder's pools (merged in)."""
     import sptag_amd
