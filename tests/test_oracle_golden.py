@@ -64,3 +64,24 @@ def test_oracle_deletes():
     ix = OrcIndex.from_arrays(vec, tstart, tnodes, graph, "L2", deleted=deleted)
     vids, _ = ix.search_batch(g["queries"], 10, 2048, nthreads=2)
     assert not np.isin(vids[vids >= 0], np.where(deleted)[0]).any()
+
+
+def test_oracle_edge_params():
+    """edge cases the reference tests cover implicitly: k larger than the
+    reachable set pads with VID=-1/MaxDist; tiny MaxCheck still returns
+    sorted results; single-query batch."""
+    g = load_golden("f32_l2_grid_ties")
+    ix = OrcIndex.load(g["index"])
+    q = g["queries"][:4]
+    vids, dists = ix.search_batch(q, 64, 1, nthreads=1)  # mc=1: tiny budget
+    for i in range(4):
+        row = vids[i]
+        valid = row[row >= 0]
+        assert len(valid) == len(set(valid.tolist()))     # no duplicates
+        dd = dists[i][row >= 0]
+        assert (dd[1:] >= dd[:-1]).all()                  # sorted
+    assert (vids >= -1).all()
+    # k=1
+    v1, d1 = ix.search_batch(q, 1, 2048, nthreads=1)
+    v64, d64 = ix.search_batch(q, 64, 2048, nthreads=1)
+    np.testing.assert_array_equal(v1[:, 0], v64[:, 0])
