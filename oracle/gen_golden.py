@@ -81,6 +81,57 @@ def gen_fixture(name, data, queries, vtype, dist, maxchecks=MAXCHECKS, k=K,
     print(f"fixture {name} done")
 
 
+def gen_fixture_ex(name, data, queries, vtype, dist, maxchecks, algo="BKT",
+                   k=K, extra=()):
+    """gen_fixture with extra builder parameters (Section.Name=Value argv,
+    reference IndexSearcher/IndexBuilder main.cpp:362-389)."""
+    d = os.path.join(GOLDEN, name)
+    idx = os.path.join(d, "index")
+    os.makedirs(d, exist_ok=True)
+    write_default(os.path.join(d, "data.bin"), data)
+    write_default(os.path.join(d, "queries.bin"), queries)
+    run([os.path.join(REF, "indexbuilder"), "-d", str(data.shape[1]),
+         "-v", vtype, "-f", "DEFAULT", "-i", os.path.join(d, "data.bin"),
+         "-o", idx, "-a", algo, "-t", "4",
+         "Index.DistCalcMethod=" + dist, *extra])
+    for mc in maxchecks:
+        run([os.path.join(REF, "indexsearcher"), "-d", str(data.shape[1]),
+             "-v", vtype, "-f", "DEFAULT", "-i", os.path.join(d, "queries.bin"),
+             "-x", idx, "-k", str(k), "-m", str(mc), "-t", "2", "-of", "1",
+             "-o", os.path.join(d, f"results_mc{mc}.bin")])
+    os.remove(os.path.join(d, "data.bin"))
+    with open(os.path.join(d, "meta.json"), "w") as f:
+        json.dump({"n": int(data.shape[0]), "dim": int(data.shape[1]),
+                   "valuetype": vtype, "distmethod": dist, "k": k, "algo": algo,
+                   "maxchecks": maxchecks, "nq": int(queries.shape[0])}, f)
+    print(f"fixture {name} done")
+
+
+def gen_iter_and_add_goldens():
+    """Iterative-search and online-add golden answers, produced by driving
+    the reference's OWN ResultIterator / AddIndex (iterprobe / addprobe
+    harnesses linked against the oracle/_ref objects)."""
+    rng = np.random.default_rng(555)
+    for name in ["f32_l2_n10k_d32", "i8_l2_n10k_d100"]:
+        d = os.path.join(GOLDEN, name)
+        meta = json.load(open(os.path.join(d, "meta.json")))
+        run([os.path.join(REF, "iterprobe"), os.path.join(d, "index"),
+             os.path.join(d, "queries.bin"), "8", "5",
+             os.path.join(d, "iter_b8_c5.bin")])
+        dim = meta["dim"]
+        if meta["valuetype"] == "Float":
+            add = (rng.random((64, dim), dtype=np.float32) * 100).astype(np.float32)
+        else:
+            add = rng.integers(-100, 101, (64, dim)).astype(np.int8)
+        write_default(os.path.join(d, "add_vectors.bin"), add)
+        out = os.path.join("/tmp", "postadd_" + name)
+        run([os.path.join(REF, "addprobe"), os.path.join(d, "index"),
+             os.path.join(d, "add_vectors.bin"), out])
+        import shutil
+        shutil.copy(os.path.join(out, "graph.bin"),
+                    os.path.join(d, "postadd_graph.bin"))
+
+
 def main():
     os.makedirs(GOLDEN, exist_ok=True)
     rng = np.random.default_rng(2016)
@@ -143,6 +194,29 @@ def main():
     queries = rng.standard_normal((100, 48)).astype(np.float32)
     gen_fixture("f32_cos_n10k_d48", data, queries, "Float", "Cosine",
                 maxchecks=[512, 2048, 8192])
+
+    # 8. builder-parameter coverage: multi-tree BKT, non-default degree,
+    #    2-tree KDT, odd int8 dim (byte-wise distance path)
+    rng2 = np.random.default_rng(777)
+    data = rng2.random((8000, 32), dtype=np.float32) * 100
+    queries = rng2.random((100, 32), dtype=np.float32) * 100
+    gen_fixture_ex("f32_l2_bkt2trees", data, queries, "Float", "L2",
+                   [512, 2048], extra=("Index.BKTNumber=2",))
+    data = rng2.random((8000, 48), dtype=np.float32)
+    queries = rng2.random((100, 48), dtype=np.float32)
+    gen_fixture_ex("f32_l2_deg16", data, queries, "Float", "L2", [512, 2048],
+                   extra=("Index.NeighborhoodSize=16",))
+    data = rng2.integers(-100, 101, (8000, 100)).astype(np.int8)
+    queries = rng2.integers(-100, 101, (100, 100)).astype(np.int8)
+    gen_fixture_ex("kdt_i8_cos_2trees", data, queries, "Int8", "Cosine",
+                   [2048, 8192], algo="KDT", extra=("Index.KDTNumber=2",))
+    rng3 = np.random.default_rng(4242)
+    data = rng3.integers(-100, 101, (8000, 37)).astype(np.int8)
+    queries = rng3.integers(-100, 101, (100, 37)).astype(np.int8)
+    gen_fixture_ex("i8_l2_odd_d37", data, queries, "Int8", "L2", [512, 2048])
+
+    # 9. iterative + online-add golden answers
+    gen_iter_and_add_goldens()
 
 
 if __name__ == "__main__":
