@@ -162,10 +162,14 @@ class AnnIndex:
     @staticmethod
     def Load(p_loaderFile):
         ix = _backend.AnnIndex.Load(p_loaderFile)
-        out = AnnIndex("BKT", "Float", ix.dim)
+        out = AnnIndex("KDT" if ix.algo == 1 else "BKT", "Float", ix.dim)
         out._vt = "Float" if ix.valuetype == _backend.VT_FLOAT else "Int8"
         out._dim = ix.dim
         out._ix = ix
+        # reflect the loaded index's metric so later Build()/introspection
+        # on this handle uses the right default
+        out._params["DistCalcMethod"] = (
+            "L2" if ix.distmethod == _backend.DM_L2 else "Cosine")
         return out
 
     def ReadyToServe(self):

@@ -161,6 +161,8 @@ int     sptag_amd_valuetype(const SptagAmdIndex* idx);
 int     sptag_amd_distmethod(const SptagAmdIndex* idx);
 int32_t sptag_amd_degree(const SptagAmdIndex* idx);
 int32_t sptag_amd_default_maxcheck(const SptagAmdIndex* idx);
+/* 0 = BKT, 1 = KDT (the loaded indexloader.ini IndexAlgoType). */
+int     sptag_amd_algo(const SptagAmdIndex* idx);
 
 /* Runtime probe: 1 if a usable HIP device exists (search will run), else 0. */
 int sptag_amd_gpu_available(void);

@@ -732,6 +732,13 @@ int32_t orc_iter_next(OrcIter* it, int32_t batch, int32_t* out_vids,
      * (WorkSpace.h:276-283): fresh m_Results, counters zeroed; queues,
      * visited set and m_relaxedMono persist. */
     int res_cap = it->max_check / 16 > batch ? it->max_check / 16 : batch;
+    if (res_cap > sp->results.cap_alloc) {
+        /* DistPriorityQueue::clear(count) reallocates when the requested
+         * count exceeds m_size (WorkSpace.h:176-183); mirror that grow. */
+        sp->results.a = (float*)realloc(sp->results.a,
+                                        sizeof(float) * (size_t)(res_cap + 1));
+        sp->results.cap_alloc = res_cap;
+    }
     sp->results.a[1] = ORC_MAXDIST;
     sp->results.length = 1;
     sp->results.count = res_cap;

@@ -75,7 +75,7 @@ def load_library():
     lib.sptag_amd_save_index.restype = ctypes.c_int
     lib.sptag_amd_save_index.argtypes = [ctypes.c_void_p, ctypes.c_char_p]
     for f in ["num_vectors", "dim", "valuetype", "distmethod", "degree",
-              "default_maxcheck"]:
+              "default_maxcheck", "algo"]:
         fn = getattr(lib, "sptag_amd_" + f)
         fn.restype = ctypes.c_int32
         fn.argtypes = [ctypes.c_void_p]
@@ -209,6 +209,11 @@ class AnnIndex:
     @property
     def distmethod(self):
         return self._lib.sptag_amd_distmethod(self._h)
+
+    @property
+    def algo(self):
+        """0 = BKT, 1 = KDT (from the loaded indexloader.ini)."""
+        return self._lib.sptag_amd_algo(self._h)
 
     @property
     def degree(self):

@@ -235,6 +235,12 @@ SptagAmdIndex* sptag_amd_create_index_kdt(int32_t n, int32_t dim, int valuetype,
     if (n <= 0 || dim <= 0 || dim > MAX_DIM || degree <= 0 || degree > MAX_DEG ||
         ntrees <= 0 || n_tree_nodes <= 0)
         return nullptr;
+    if (valuetype == SPTAG_AMD_VT_INT8 && (int64_t)dim * 254 * 254 >= (1ll << 24)) {
+        /* same exact-float-range guard as the BKT create path: beyond it the
+         * int8 distances stop being exact integers in float. */
+        fprintf(stderr, "sptag_amd: int8 dim %d exceeds exact-float range\n", dim);
+        return nullptr;
+    }
     auto* ix = new SptagAmdIndex();
     ix->device = device;
     ix->algo = ALGO_KDT;
@@ -538,6 +544,20 @@ static int search_device_core(SptagAmdIndex* ix, const void* d_q, int32_t nq,
         for (int32_t i = 0; i < gq_n; i++) {
             ix->last_checked += stats[2 * (size_t)i];
             ix->last_popped += stats[2 * (size_t)i + 1];
+        }
+        /* a query can still overflow at reference capacities only through
+         * visited-table saturation (probe give-up) — that would be a silent
+         * result truncation, so it is an error, not a degradation. */
+        std::vector<int32_t> oflow2(gq_n);
+        HIP_OR_FAIL(hipMemcpy(oflow2.data(), ix->d_oflow, (size_t)gq_n * 4,
+                              hipMemcpyDeviceToHost), SPTAG_AMD_ERR_NOGPU);
+        for (int32_t i = 0; i < gq_n; i++) {
+            if (oflow2[i]) {
+                fprintf(stderr,
+                        "sptag_amd: query overflowed at reference capacities "
+                        "(visited-table saturation)\n");
+                return SPTAG_AMD_ERR_INTERNAL;
+            }
         }
     }
     return SPTAG_AMD_OK;
@@ -1154,5 +1174,6 @@ int sptag_amd_distmethod(const SptagAmdIndex* ix) { return ix ? ix->dm : -1; }
 int32_t sptag_amd_degree(const SptagAmdIndex* ix) { return ix ? ix->deg : -1; }
 int32_t sptag_amd_default_maxcheck(const SptagAmdIndex* ix)
 { return ix ? ix->default_maxcheck : -1; }
+int sptag_amd_algo(const SptagAmdIndex* ix) { return ix ? ix->algo : -1; }
 
 }  /* extern "C" */
