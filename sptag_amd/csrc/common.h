@@ -39,6 +39,12 @@ struct SearchCfg {
     int32_t search_deleted;      /* searchDeleted (dispatch bit 2) */
     int32_t ng_cap, spt_cap, dpq_cap;  /* heap capacities (entries) */
     int32_t vcap;                      /* visited table slots (pow2) */
+    int32_t spec_flags;  /* perf-only speculation (identical results):
+                            bit0 = warm neighbor vector lines during the
+                                   visited-CAS round;
+                            bit1 = next-pop lookahead (read next frontier
+                                   top's adjacency, read-only visited probe,
+                                   start unvisited vector loads) */
 };
 
 /* per-launch buffers */

@@ -416,6 +416,8 @@ static int search_device_core(SptagAmdIndex* ix, const void* d_q, int32_t nq,
     if (const char* e = getenv("SPTAG_AMD_NG_CAP"))   /* perf experiments */
         cfg.ng_cap = std::max(256, atoi(e));
     cfg.spt_cap = 4096;
+    cfg.spec_flags = 3;   /* perf-only speculation; identical results */
+    if (const char* e = getenv("SPTAG_AMD_SPEC")) cfg.spec_flags = atoi(e);
 
     int lds_limit = 64 * 1024;
     (void)hipDeviceGetAttribute(&lds_limit, hipDeviceAttributeMaxSharedMemoryPerBlock,
@@ -736,6 +738,8 @@ int sptag_amd_iter_next(SptagAmdIterBatch* it, int32_t batch,
     cfg.vcap = it->vcap;
     cfg.ng_cap = it->ng_cap;
     cfg.spt_cap = it->spt_cap;
+    cfg.spec_flags = 3;
+    if (const char* e = getenv("SPTAG_AMD_SPEC")) cfg.spec_flags = atoi(e);
 
     IterBufs ib;
     ib.queries = it->d_q;

@@ -277,7 +277,13 @@ DEV float ref_dist_grp_f32(const float* __restrict__ q,
 }
 
 /* int8: exact integer accumulation (order-free; partials exact in the
- * reference's float lanes for dim*254^2 < 2^24 — guarded host-side). */
+ * reference's float lanes for dim*254^2 < 2^24 — guarded host-side).
+ * The dword path runs on the CDNA4 int8 dot units (v_dot4c_i32_i8 via
+ * __builtin_amdgcn_sdot4) — the gfx950 analog of the reference's own GPU
+ * __dp4a int8 path (cuda/Distance.hxx:84-101). Integer dots are exact and
+ * order-free, so results stay bit-identical to the byte-wise form:
+ * L2 uses sum (xa-xb)^2 = dot(a,a) - 2 dot(a,b) + dot(b,b), every term
+ * < 2^25 at the guarded dims. */
 template <int DM>
 DEV float ref_dist_grp_i8(const int8_t* __restrict__ q,
                           const int8_t* __restrict__ v, int d)
@@ -288,14 +294,18 @@ DEV float ref_dist_grp_i8(const int8_t* __restrict__ q,
         const int W = d >> 2;
         const int32_t* qw = (const int32_t*)q;
         const int32_t* vw = (const int32_t*)v;
-        for (int w = g; w < W; w += 16) {
-            int32_t a = qw[w], b = vw[w];
-#pragma unroll
-            for (int byte = 0; byte < 4; byte++) {
-                int xa = (int)(int8_t)(a >> (8 * byte));
-                int xb = (int)(int8_t)(b >> (8 * byte));
-                s += (DM == DM_L2) ? (xa - xb) * (xa - xb) : xa * xb;
+        if (DM == DM_COSINE) {
+            for (int w = g; w < W; w += 16)
+                s = __builtin_amdgcn_sdot4(qw[w], vw[w], s, false);
+        } else {
+            int aa = 0, ab = 0, bb = 0;
+            for (int w = g; w < W; w += 16) {
+                int32_t a = qw[w], b = vw[w];
+                aa = __builtin_amdgcn_sdot4(a, a, aa, false);
+                ab = __builtin_amdgcn_sdot4(a, b, ab, false);
+                bb = __builtin_amdgcn_sdot4(b, b, bb, false);
             }
+            s = aa - 2 * ab + bb;
         }
     } else {
         for (int i = g; i < d; i += 16) {
@@ -586,13 +596,16 @@ void bkt_search_kernel(DevIndex di, SearchCfg cfg, SearchBufs bufs)
 
     const int deg = di.deg;
     const int checkPos = deg - 1;
+    const int nline = ((int)((unsigned)di.dim * sizeof(T)) + 127) >> 7;
     int popped = 0;
+    uint32_t spec_sink = 0;   /* keeps lookahead loads alive across barriers */
 
     for (;;) {
         __syncthreads();
         if (ss->ng_count <= 0 || ss->terminate || ss->oflow) break;
         if (lane == 0) ss->popped = ndheap_pop(c.ng, &ss->ng_count);
         __syncthreads();
+        asm volatile("" :: "v"(spec_sink));   /* lookahead loads land here */
         popped++;
         NodeDist gnode = ss->popped;
         const int32_t* row = di.graph + (size_t)gnode.node * deg;
@@ -639,11 +652,24 @@ void bkt_search_kernel(DevIndex di, SearchCfg cfg, SearchBufs bufs)
         if (ss->terminate) break;
 
         /* neighbor expansion (BKTIndex.cpp:333-345): compact the unvisited
-         * neighbors (row order preserved) and stage their distances. */
+         * neighbors (row order preserved) and stage their distances.
+         * spec bit0: while the visited-CAS round is in flight, warm every
+         * neighbor's vector lines (one byte per 128B line) — the unvisited
+         * ones are exactly what stage_dists reads next, so the two
+         * dependent HBM round trips (CAS, vector gather) overlap; the
+         * re-visited rows were read recently and mostly hit L2/L3. */
         int already = 1;
-        if (lane < firstneg)
+        uint32_t tsink = 0;
+        if (lane < firstneg) {
+            if (cfg.spec_flags & 1) {
+                const char* vp = (const char*)vec_at<T>(di, nn);
+                for (int l = 0; l < nline; l++)
+                    tsink += (uint32_t)(uint8_t)vp[(size_t)l << 7];
+            }
             already = visited_test_insert(c.vtab, c.vmask, nn, &ss->oflow);
+        }
         uint64_t candm = __ballot(lane < firstneg && !already);
+        asm volatile("" :: "v"(tsink));
         int ncand = __popcll(candm);
         if ((candm >> lane) & 1) {
             int pos = __popcll(candm & ((1ull << lane) - 1));
@@ -665,16 +691,27 @@ void bkt_search_kernel(DevIndex di, SearchCfg cfg, SearchBufs bufs)
         __syncthreads();
         if (ss->want_tree)
             search_trees_dev<T, DM>(c, cfg.other_pivots + ss->checked);
-        /* warm L1/L2 with the next pop's adjacency row (speculative,
-         * perf-only: the top rarely changes before the next iteration). */
-        {
-            int32_t nxt = ss->popped.node;   /* broadcast slot reused below */
+        /* spec bit1: lookahead on the next pop (perf-only — the frontier
+         * top rarely changes between here and the next iteration's pop).
+         * Read its adjacency row, probe the visited table READ-ONLY (first
+         * slot; stale/partial answers only cost an extra touch), and start
+         * the unvisited neighbors' vector loads. The loads stay in flight
+         * across the barrier — spec_sink is consumed after the next pop. */
+        if (cfg.spec_flags & 2) {
             if (lane == 0) ss->popped = ndheap_top(c.ng, ss->ng_count);
             __syncthreads();
-            nxt = ss->popped.node;
+            int32_t nxt = ss->popped.node;
             if (nxt >= 0 && lane < deg) {
                 int32_t t = di.graph[(size_t)nxt * deg + lane];
-                asm volatile("" :: "v"(t));
+                if (t >= 0) {
+                    uint32_t key = (uint32_t)(t + 1);
+                    uint32_t h = (key * 2654435761u) & c.vmask;
+                    if (c.vtab[h] != (int32_t)key) {
+                        const char* vp = (const char*)vec_at<T>(di, t);
+                        for (int l = 0; l < nline; l++)
+                            spec_sink += (uint32_t)(uint8_t)vp[(size_t)l << 7];
+                    }
+                }
             }
         }
     }
@@ -815,10 +852,21 @@ void bkt_iter_kernel(DevIndex di, SearchCfg cfg, IterBufs ib, int batch)
             }
         }
 
+        /* spec bit0 (see bkt kernel): warm neighbor vector lines during
+         * the visited-CAS round. */
         int already = 1;
-        if (lane < firstneg)
+        uint32_t tsink = 0;
+        if (lane < firstneg) {
+            if (cfg.spec_flags & 1) {
+                const char* vp = (const char*)vec_at<T>(di, nn);
+                const int nline = ((int)((unsigned)di.dim * sizeof(T)) + 127) >> 7;
+                for (int l = 0; l < nline; l++)
+                    tsink += (uint32_t)(uint8_t)vp[(size_t)l << 7];
+            }
             already = visited_test_insert(c.vtab, c.vmask, nn, &ss->oflow);
+        }
         uint64_t candm = __ballot(lane < firstneg && !already);
+        asm volatile("" :: "v"(tsink));
         int ncand = __popcll(candm);
         if ((candm >> lane) & 1) {
             int pos = __popcll(candm & ((1ull << lane) - 1));
@@ -982,13 +1030,16 @@ void kdt_search_kernel(DevIndex di, SearchCfg cfg, SearchBufs bufs)
     kdt_search_trees_dev<T, DM>(c, cfg.init_pivots);
 
     const int deg = di.deg;
+    const int nline = ((int)((unsigned)di.dim * sizeof(T)) + 127) >> 7;
     int popped = 0;
+    uint32_t spec_sink = 0;
 
     for (;;) {
         __syncthreads();
         if (ss->ng_count <= 0 || ss->terminate || ss->oflow) break;
         if (lane == 0) ss->popped = ndheap_pop(c.ng, &ss->ng_count);
         __syncthreads();
+        asm volatile("" :: "v"(spec_sink));
         popped++;
         NodeDist gnode = ss->popped;
         const int32_t* row = di.graph + (size_t)gnode.node * deg;
@@ -1010,10 +1061,20 @@ void kdt_search_kernel(DevIndex di, SearchCfg cfg, SearchBufs bufs)
         if (ss->terminate) break;
         float upper_bound = ss->fbcast;
 
+        /* spec bit0 (see bkt kernel): warm neighbor vector lines during
+         * the visited-CAS round. */
         int already = 1;
-        if (lane < firstneg)
+        uint32_t tsink = 0;
+        if (lane < firstneg) {
+            if (cfg.spec_flags & 1) {
+                const char* vp = (const char*)vec_at<T>(di, nn);
+                for (int l = 0; l < nline; l++)
+                    tsink += (uint32_t)(uint8_t)vp[(size_t)l << 7];
+            }
             already = visited_test_insert(c.vtab, c.vmask, nn, &ss->oflow);
+        }
         uint64_t candm = __ballot(lane < firstneg && !already);
+        asm volatile("" :: "v"(tsink));
         int ncand = __popcll(candm);
         if ((candm >> lane) & 1) {
             int pos = __popcll(candm & ((1ull << lane) - 1));
@@ -1046,6 +1107,24 @@ void kdt_search_kernel(DevIndex di, SearchCfg cfg, SearchBufs bufs)
         if (ss->break_flag) break;
         if (ss->want_tree)
             kdt_search_trees_dev<T, DM>(c, cfg.other_pivots + ss->checked);
+        /* spec bit1 (see bkt kernel): next-pop lookahead. */
+        if (cfg.spec_flags & 2) {
+            if (lane == 0) ss->popped = ndheap_top(c.ng, ss->ng_count);
+            __syncthreads();
+            int32_t nxt = ss->popped.node;
+            if (nxt >= 0 && lane < deg) {
+                int32_t t = di.graph[(size_t)nxt * deg + lane];
+                if (t >= 0) {
+                    uint32_t key = (uint32_t)(t + 1);
+                    uint32_t h = (key * 2654435761u) & c.vmask;
+                    if (c.vtab[h] != (int32_t)key) {
+                        const char* vp = (const char*)vec_at<T>(di, t);
+                        for (int l = 0; l < nline; l++)
+                            spec_sink += (uint32_t)(uint8_t)vp[(size_t)l << 7];
+                    }
+                }
+            }
+        }
     }
     __syncthreads();
 
