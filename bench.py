@@ -58,7 +58,21 @@ CONFIGS = {
     "bkt_100m_d100_i8_cos": dict(n=100_000_000, d=100, dtype="i8",
                                  metric="Cosine", nq=10_000, k=10,
                                  ncenters=16384, sigma=30.0, cand=128,
-                                 ntrees=4, refine=0, srefine=1),
+                                 ntrees=4, refine=2, srefine=1,
+                                 srefine_k=256, srefine_mc=2048),
+    # 30M validation scale for the config-#3 build recipe
+    "bkt_30m_d100_i8_cos": dict(n=30_000_000, d=100, dtype="i8",
+                                metric="Cosine", nq=10_000, k=10,
+                                ncenters=16384, sigma=30.0, cand=128,
+                                ntrees=4, refine=2, srefine=1,
+                                srefine_k=256, srefine_mc=2048),
+    # BASELINE.json configs[4] — 1B int8 L2, meant for --gpus 8 (125M rows
+    # per shard; per-shard recipe = the config-#3 recipe)
+    "bkt_1b_d100_i8_l2": dict(n=1_000_000_000, d=100, dtype="i8",
+                              metric="L2", nq=10_000, k=10,
+                              ncenters=65536, sigma=30.0, cand=128,
+                              ntrees=4, refine=2, srefine=1,
+                              srefine_k=256, srefine_mc=2048),
     # BASELINE.json configs[3] — KDT, embedding shape. Note: the KDT
     # algorithm's no-better-propagation termination caps recall on this
     # data family at ~0.93-0.94 for the REFERENCE implementation as well
@@ -265,7 +279,13 @@ def main():
     ap.add_argument("--ntrees", type=int, default=4)
     ap.add_argument("--cand", type=int, default=0)
     ap.add_argument("--ncenters", type=int, default=0)
-    ap.add_argument("--refine", type=int, default=0)
+    ap.add_argument("--refine", type=int, default=-1,
+                    help="NN-descent refine rounds (override config)")
+    ap.add_argument("--srefine", type=int, default=-1,
+                    help="search-refine rounds (override config)")
+    ap.add_argument("--srefine-k", type=int, default=0)
+    ap.add_argument("--srefine-mc", type=int, default=0)
+    ap.add_argument("--fill-pruned", type=int, default=-1)
     args = ap.parse_args()
 
     import torch
@@ -296,10 +316,16 @@ def main():
     x_np = x.cpu().numpy()
     arrays = build_index_arrays(
         x_np, cfg["metric"], algo=cfg.get("algo", "BKT"),
-        cand=args.cand or cfg.get("cand", 256), kdt_trees=2,
+        cand=args.cand or cfg.get("cand", 256),
+        kdt_trees=cfg.get("kdt_trees", 2),
         ntrees=cfg.get("ntrees", args.ntrees),
-        refine_rounds=cfg.get("refine", args.refine),
-        search_refine_rounds=cfg.get("srefine", 0),
+        refine_rounds=args.refine if args.refine >= 0 else cfg.get("refine", 0),
+        search_refine_rounds=(args.srefine if args.srefine >= 0
+                              else cfg.get("srefine", 0)),
+        srefine_k=args.srefine_k or cfg.get("srefine_k", 512),
+        srefine_mc=args.srefine_mc or cfg.get("srefine_mc", 8192),
+        fill_pruned=bool(args.fill_pruned if args.fill_pruned >= 0
+                         else cfg.get("fill_pruned", False)),
         device=device, normalized=False,
         verbose=(rank == 0))
     log(rank, f"index built ({time.time()-t0:.1f}s)")

@@ -585,8 +585,14 @@ def build_rng_graph(vectors, *, degree=32, ntrees=4, tpt_leaf=1000, cand=256,
     return graph, ids, dst
 
 
-def _rng_prune(xf, cid, cdd, degree, rng_factor, device):
-    """RNG prune one chunk of candidate lists (ascending by dist)."""
+def _rng_prune(xf, cid, cdd, degree, rng_factor, device, fill_pruned=False):
+    """RNG prune one chunk of candidate lists (ascending by dist).
+
+    fill_pruned: after the RNG rule fills what it can, pad remaining degree
+    slots with the nearest REJECTED candidates (build-quality knob for
+    billion-scale pools, cf. HNSW keepPrunedConnections; the reference's
+    RebuildNeighbors pads -1 — its CEF=1000 pools rarely leave slots empty,
+    shallow pools at 100M+ do)."""
     C = cid.shape[1]
     cvec = xf[cid.clamp(min=0).long()]
     csq = (cvec * cvec).sum(-1)
@@ -599,6 +605,12 @@ def _rng_prune(xf, cid, cdd, degree, rng_factor, device):
         good = valid[:, j] & ~viol.any(1) & (count < degree)
         acc[:, j] = good
         count += good.int()
+    if fill_pruned:
+        # fill leftover slots with the nearest valid rejected candidates
+        free = (degree - count).clamp(min=0)
+        rej = valid & ~acc
+        rej_rank = torch.cumsum(rej.int(), dim=1)
+        acc = acc | (rej & (rej_rank <= free[:, None]))
     out = torch.full((cid.shape[0], degree), -1, dtype=torch.int32, device=device)
     pos = (torch.cumsum(acc.int(), dim=1) - 1).clamp(min=0)
     rows = torch.nonzero(acc, as_tuple=True)
@@ -628,22 +640,23 @@ def refine_graph(vectors, graph, cand_ids, cand_dst, *, degree=32, cand=256,
         point_chunk = _prune_chunk(cand, xf.shape[1])
 
     for r in range(rounds):
-        # reverse edges (sampled): every edge (i -> j) proposes i to j
-        src = self_ids.repeat_interleave(g.shape[1])
-        dstv = g.reshape(-1)
-        keep = dstv >= 0
-        src, dstv = src[keep], dstv[keep]
-        order = torch.argsort(dstv, stable=True)
-        src, dstv = src[order], dstv[order]
-        cnt = torch.bincount(dstv.long(), minlength=n)
-        start = torch.cumsum(cnt, 0) - cnt
+        # reverse edges (sampled): every edge (i -> j) proposes i to j.
+        # Slot by a per-round hash of the source and let collisions drop
+        # (last-write-wins scatter) — O(E) with no 32n-element sort, which
+        # at 100M rows would blow past cub's 2^31 sort limit and ~65 GB of
+        # transients. NN-descent only needs a reverse SAMPLE per node.
         rcap = 16
-        pos = torch.arange(rcap, device=device)
-        ridx = (start[:, None] + pos[None, :]).clamp(max=max(len(src) - 1, 0))
-        rmask = pos[None, :] < cnt[:, None]
         rev = torch.full((n, rcap), -1, dtype=torch.int32, device=device)
-        if len(src):
-            rev[rmask] = src[ridx[rmask]]
+        edge_chunk = 64_000_000
+        deg = g.shape[1]
+        for es in range(0, n, max(1, edge_chunk // deg)):
+            ee = min(n, es + max(1, edge_chunk // deg))
+            srcc = self_ids[es:ee].repeat_interleave(deg)
+            dstc = g[es:ee].reshape(-1).long()
+            keep = dstc >= 0
+            srcc, dstc = srcc[keep], dstc[keep]
+            slot = ((srcc.long() * 2654435761) + r * 97) % rcap
+            rev[dstc, slot] = srcc
         for s in range(0, n, point_chunk):
             e = min(n, s + point_chunk)
             B = e - s
@@ -675,7 +688,8 @@ def _exact_l2(xf, q_rows, c_ids):
 def refine_via_search(vectors_t, tree_start, tree_nodes, graph, cand_ids,
                       cand_dst, distmethod, *, algo="BKT", degree=32,
                       cand=256, rounds=1, k=512, max_check=8192,
-                      rng_factor=1.0, chunk=1_000_000, verbose=False):
+                      rng_factor=1.0, chunk=0, fill_pruned=False,
+                      verbose=False):
     """The reference's own refinement recipe (NeighborhoodGraph.h:460-560
     RefineGraph: every node re-searches the CURRENT index and its edges are
     RNG-rebuilt from the results) — run on the PRODUCT GPU searcher via the
@@ -691,6 +705,13 @@ def refine_via_search(vectors_t, tree_start, tree_nodes, graph, cand_ids,
     n = vectors_t.shape[0]
     xf = vectors_t if vectors_t.dtype == torch.float32 else vectors_t.float()
     self_ids = torch.arange(n, device=device, dtype=torch.int32)
+    if chunk <= 0:
+        # the searcher's per-query visited table is 4*next_pow2(4*max_check)
+        # bytes; keep the whole-batch scratch ~8 GB
+        vcap = 1
+        while vcap < max(4096, max_check * 4):
+            vcap <<= 1
+        chunk = min(1_000_000, max(65_536, int(8e9 // (vcap * 4))))
     x_np = vectors_t.cpu().numpy()
     graph_t = torch.as_tensor(graph, device=device)         if not torch.is_tensor(graph) else graph
     # search distances arrive in the index metric; the builder's pools are
@@ -719,7 +740,8 @@ def refine_via_search(vectors_t, tree_start, tree_nodes, graph, cand_ids,
                 cand_ids[s0:e0], cand_dst[s0:e0], sv, sd, cand,
                 self_ids[s0:e0])
             graph_t[s0:e0] = _rng_prune(xf, cand_ids[s0:e0], cand_dst[s0:e0],
-                                        degree, rng_factor, device)
+                                        degree, rng_factor, device,
+                                        fill_pruned=fill_pruned)
             if verbose and (s0 // chunk) % 10 == 0:
                 print(f"  search-refine round {r + 1}: {e0}/{n}")
         del ix, d_vids, d_dists
@@ -732,6 +754,7 @@ def refine_via_search(vectors_t, tree_start, tree_nodes, graph, cand_ids,
 def build_index_arrays(vectors, distmethod, *, algo="BKT", degree=32, ntrees=4,
                        tpt_leaf=1000, cand=256, kmeans_k=32, leaf_size=32,
                        refine_rounds=0, search_refine_rounds=0, kdt_trees=1,
+                       srefine_k=512, srefine_mc=8192, fill_pruned=False,
                        seed=2016, device=None, normalized=False,
                        verbose=False):
     """Full build: returns dict(vectors, tree_start, tree_nodes, graph) ready
@@ -760,7 +783,8 @@ def build_index_arrays(vectors, distmethod, *, algo="BKT", degree=32, ntrees=4,
         xt = torch.as_tensor(vectors, device=device or _dev())
         graph, cids, cdst = refine_via_search(
             xt, tree_start, tree_nodes, graph, cids, cdst, distmethod,
-            algo=algo, degree=degree, cand=cand,
+            algo=algo, degree=degree, cand=cand, k=srefine_k,
+            max_check=srefine_mc, fill_pruned=fill_pruned,
             rounds=search_refine_rounds, verbose=verbose)
         del xt
     graph = graph.cpu().numpy() if torch.is_tensor(graph) else graph

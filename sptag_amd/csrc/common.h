@@ -45,7 +45,15 @@ struct SearchCfg {
                             bit1 = next-pop lookahead (read next frontier
                                    top's adjacency, read-only visited probe,
                                    start unvisited vector loads) */
+    int32_t prof;        /* 1 = per-phase cycle breakdown into stats
+                            (stride PROF_STATS int32 per query; lane-0
+                            clock64 marks — one wave per WG makes lane-0
+                            spans wave-accurate). Diagnostic only. */
 };
+
+enum { PROF_STATS = 16 };  /* stats stride (int32) in prof mode; slots:
+    0 checked, 1 popped, 2 seed, 3 pop, 4 row, 5 serial, 6 cas, 7 dist,
+    8 insert, 9 tree, 10 spec, 11 epilogue (cycles) */
 
 /* per-launch buffers */
 struct SearchBufs {

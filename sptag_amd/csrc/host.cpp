@@ -416,8 +416,14 @@ static int search_device_core(SptagAmdIndex* ix, const void* d_q, int32_t nq,
     if (const char* e = getenv("SPTAG_AMD_NG_CAP"))   /* perf experiments */
         cfg.ng_cap = std::max(256, atoi(e));
     cfg.spt_cap = 4096;
-    cfg.spec_flags = 3;   /* perf-only speculation; identical results */
+    /* measured on MI355X (gpurun_out/ab_v5.txt): both speculation forms
+     * LOSE throughput at 10M/mc2048 (745k QPS off vs 622k on) — the extra
+     * touch loads consume request bandwidth the resident waves already
+     * saturate. Off by default; kept for experiments. */
+    cfg.spec_flags = 0;
     if (const char* e = getenv("SPTAG_AMD_SPEC")) cfg.spec_flags = atoi(e);
+    cfg.prof = getenv("SPTAG_AMD_PROF") ? 1 : 0;   /* phase-cycle diagnostic */
+    const int sstride = cfg.prof ? PROF_STATS : 2;
 
     int lds_limit = 64 * 1024;
     (void)hipDeviceGetAttribute(&lds_limit, hipDeviceAttributeMaxSharedMemoryPerBlock,
@@ -428,7 +434,8 @@ static int search_device_core(SptagAmdIndex* ix, const void* d_q, int32_t nq,
                    (size_t)nq * cfg.vcap * 4) != SPTAG_AMD_OK) return SPTAG_AMD_ERR_OOM;
     if (ensure_cap((void**)&ix->d_oflow, &ix->oflow_cap, (size_t)nq * 4) != SPTAG_AMD_OK)
         return SPTAG_AMD_ERR_OOM;
-    if (ensure_cap((void**)&ix->d_stats, &ix->stats_cap, (size_t)nq * 8) != SPTAG_AMD_OK)
+    if (ensure_cap((void**)&ix->d_stats, &ix->stats_cap,
+                   (size_t)nq * sstride * 4) != SPTAG_AMD_OK)
         return SPTAG_AMD_ERR_OOM;
     if (!ix->ev0) { (void)hipEventCreate(&ix->ev0); (void)hipEventCreate(&ix->ev1); }
 
@@ -449,7 +456,7 @@ static int search_device_core(SptagAmdIndex* ix, const void* d_q, int32_t nq,
     ix->last_kernel_ms = 0;
     ix->last_checked = 0;
     ix->last_popped = 0;
-    std::vector<int32_t> stats(2 * (size_t)nq);
+    std::vector<int32_t> stats((size_t)sstride * nq);
     std::vector<int32_t> redo;
     bool all_global = !lds_ok;
     if (lds_ok) {
@@ -469,15 +476,32 @@ static int search_device_core(SptagAmdIndex* ix, const void* d_q, int32_t nq,
         std::vector<int32_t> oflow(nq);
         HIP_OR_FAIL(hipMemcpy(oflow.data(), ix->d_oflow, (size_t)nq * 4,
                               hipMemcpyDeviceToHost), SPTAG_AMD_ERR_NOGPU);
-        HIP_OR_FAIL(hipMemcpy(stats.data(), ix->d_stats, (size_t)nq * 8,
+        HIP_OR_FAIL(hipMemcpy(stats.data(), ix->d_stats,
+                              (size_t)nq * sstride * 4,
                               hipMemcpyDeviceToHost), SPTAG_AMD_ERR_NOGPU);
         for (int32_t i = 0; i < nq; i++) {
             if (oflow[i]) {
                 redo.push_back(i);
             } else {
-                ix->last_checked += stats[2 * (size_t)i];
-                ix->last_popped += stats[2 * (size_t)i + 1];
+                ix->last_checked += stats[(size_t)sstride * i];
+                ix->last_popped += stats[(size_t)sstride * i + 1];
             }
+        }
+        if (cfg.prof && ix->algo == ALGO_BKT) {
+            static const char* pname[10] = {
+                "seed", "pop", "row", "serial", "cas",
+                "dist", "insert", "tree", "spec", "sort"};
+            double tot[10] = {};
+            for (int32_t i = 0; i < nq; i++)
+                for (int p = 0; p < 10; p++)
+                    tot[p] += (double)stats[(size_t)sstride * i + 2 + p];
+            double all = 0;
+            for (int p = 0; p < 10; p++) all += tot[p];
+            fprintf(stderr, "sptag_amd PROF (sum of per-query wave cycles, "
+                            "%d queries, kernel %.2f ms):\n", nq, ms);
+            for (int p = 0; p < 10; p++)
+                fprintf(stderr, "  %-7s %14.0f  (%5.1f%%)\n", pname[p], tot[p],
+                        all > 0 ? 100.0 * tot[p] / all : 0.0);
         }
         if (!redo.empty() && getenv("SPTAG_AMD_DEBUG"))
             fprintf(stderr, "sptag_amd: %zu/%d queries overflowed (rerun)\n",
@@ -541,11 +565,12 @@ static int search_device_core(SptagAmdIndex* ix, const void* d_q, int32_t nq,
         float ms = 0;
         (void)hipEventElapsedTime(&ms, ix->ev0, ix->ev1);
         ix->last_kernel_ms += ms;
-        HIP_OR_FAIL(hipMemcpy(stats.data(), ix->d_stats, (size_t)gq_n * 8,
+        HIP_OR_FAIL(hipMemcpy(stats.data(), ix->d_stats,
+                              (size_t)gq_n * sstride * 4,
                               hipMemcpyDeviceToHost), SPTAG_AMD_ERR_NOGPU);
         for (int32_t i = 0; i < gq_n; i++) {
-            ix->last_checked += stats[2 * (size_t)i];
-            ix->last_popped += stats[2 * (size_t)i + 1];
+            ix->last_checked += stats[(size_t)sstride * i];
+            ix->last_popped += stats[(size_t)sstride * i + 1];
         }
         /* a query can still overflow at reference capacities only through
          * visited-table saturation (probe give-up) — that would be a silent
@@ -738,8 +763,9 @@ int sptag_amd_iter_next(SptagAmdIterBatch* it, int32_t batch,
     cfg.vcap = it->vcap;
     cfg.ng_cap = it->ng_cap;
     cfg.spt_cap = it->spt_cap;
-    cfg.spec_flags = 3;
+    cfg.spec_flags = 0;
     if (const char* e = getenv("SPTAG_AMD_SPEC")) cfg.spec_flags = atoi(e);
+    cfg.prof = 0;
 
     IterBufs ib;
     ib.queries = it->d_q;

@@ -541,13 +541,22 @@ DEV void init_search_trees_dev(QCtx<T>& c)
 #define SPTAG_LB_WAVES 1   /* min waves/SIMD hint; raise to cap VGPRs */
 #endif
 
-template <typename T, int DM, bool LDSHEAP>
+/* Per-phase cycle marks, diagnostic builds only (PROF template arg).
+ * One wave per workgroup, so lane-0 clock64 spans are wave-accurate. */
+#define PMARK() do { if (PROF && lane == 0) pmark = clock64(); } while (0)
+#define PACC(i) do { if (PROF && lane == 0) { \
+        uint64_t now_ = clock64(); pc[i] += now_ - pmark; pmark = now_; } \
+    } while (0)
+
+template <typename T, int DM, bool LDSHEAP, bool PROF = false>
 __global__ __launch_bounds__(64, SPTAG_LB_WAVES)
 void bkt_search_kernel(DevIndex di, SearchCfg cfg, SearchBufs bufs)
 {
     const int q = blockIdx.x;
     if (q >= cfg.nq) return;
     const int lane = threadIdx.x;
+    uint64_t pc[10] = {};
+    uint64_t pmark = 0;
 
     extern __shared__ char smem[];
     size_t off = 0;
@@ -590,9 +599,11 @@ void bkt_search_kernel(DevIndex di, SearchCfg cfg, SearchBufs bufs)
               bufs.visited + (size_t)q * cfg.vcap, (uint32_t)(cfg.vcap - 1),
               ss, lane};
 
+    PMARK();
     init_search_trees_dev<T, DM>(c);
     __syncthreads();
     search_trees_dev<T, DM>(c, cfg.init_pivots);
+    PACC(0);
 
     const int deg = di.deg;
     const int checkPos = deg - 1;
@@ -603,8 +614,10 @@ void bkt_search_kernel(DevIndex di, SearchCfg cfg, SearchBufs bufs)
     for (;;) {
         __syncthreads();
         if (ss->ng_count <= 0 || ss->terminate || ss->oflow) break;
+        PMARK();
         if (lane == 0) ss->popped = ndheap_pop(c.ng, &ss->ng_count);
         __syncthreads();
+        PACC(1);
         asm volatile("" :: "v"(spec_sink));   /* lookahead loads land here */
         popped++;
         NodeDist gnode = ss->popped;
@@ -613,6 +626,7 @@ void bkt_search_kernel(DevIndex di, SearchCfg cfg, SearchBufs bufs)
         uint64_t negm = __ballot(lane >= deg || nn < 0);
         int firstneg = negm ? (int)__builtin_ctzll(negm) : 64;
         int32_t checkNode = __shfl(nn, checkPos);
+        PACC(2);
 
         if (lane == 0) {
             /* BKTIndex.cpp:290-331: result/termination block. The dispatch
@@ -649,6 +663,7 @@ void bkt_search_kernel(DevIndex di, SearchCfg cfg, SearchBufs bufs)
             }
         }
         __syncthreads();
+        PACC(3);
         if (ss->terminate) break;
 
         /* neighbor expansion (BKTIndex.cpp:333-345): compact the unvisited
@@ -676,7 +691,9 @@ void bkt_search_kernel(DevIndex di, SearchCfg cfg, SearchBufs bufs)
             istage[pos] = nn;
         }
         __syncthreads();
+        PACC(4);
         stage_dists<T, DM>(c, ncand);
+        PACC(5);
         if (lane == 0) {
             for (int r = 0; r < ncand; r++) {
                 float dv = dstage[r];
@@ -689,8 +706,10 @@ void bkt_search_kernel(DevIndex di, SearchCfg cfg, SearchBufs bufs)
                              ndheap_top(c.spt, ss->spt_count).distance);
         }
         __syncthreads();
+        PACC(6);
         if (ss->want_tree)
             search_trees_dev<T, DM>(c, cfg.other_pivots + ss->checked);
+        PACC(7);
         /* spec bit1: lookahead on the next pop (perf-only — the frontier
          * top rarely changes between here and the next iteration's pop).
          * Read its adjacency row, probe the visited table READ-ONLY (first
@@ -714,16 +733,28 @@ void bkt_search_kernel(DevIndex di, SearchCfg cfg, SearchBufs bufs)
                 }
             }
         }
+        PACC(8);
     }
     __syncthreads();
 
     if (lane == 0) {
         bufs.oflow[q] = ss->oflow;
-        if (bufs.stats) {
-            bufs.stats[(size_t)q * 2 + 0] = ss->checked;
-            bufs.stats[(size_t)q * 2 + 1] = popped;
-        }
+        PMARK();
         qrs_sort(qrs, cfg.k);
+        PACC(9);
+        if (bufs.stats) {
+            if (PROF) {
+                int32_t* s = bufs.stats + (size_t)q * PROF_STATS;
+                s[0] = ss->checked;
+                s[1] = popped;
+#pragma unroll
+                for (int i = 0; i < 10; i++) s[2 + i] = (int32_t)pc[i];
+                s[12] = s[13] = s[14] = s[15] = 0;
+            } else {
+                bufs.stats[(size_t)q * 2 + 0] = ss->checked;
+                bufs.stats[(size_t)q * 2 + 1] = popped;
+            }
+        }
     }
     __syncthreads();
     for (int i = lane; i < cfg.k; i += 64) {
@@ -1262,9 +1293,12 @@ static int launch_one(const DevIndex& di, const SearchCfg& cfg,
     if (di.algo == ALGO_KDT)
         hipLaunchKernelGGL((kdt_search_kernel<T, DM, LDSHEAP>), grid, block, lds,
                            stream, di, cfg, bufs);
+    else if (cfg.prof)
+        hipLaunchKernelGGL((bkt_search_kernel<T, DM, LDSHEAP, true>), grid,
+                           block, lds, stream, di, cfg, bufs);
     else
-        hipLaunchKernelGGL((bkt_search_kernel<T, DM, LDSHEAP>), grid, block, lds,
-                           stream, di, cfg, bufs);
+        hipLaunchKernelGGL((bkt_search_kernel<T, DM, LDSHEAP, false>), grid,
+                           block, lds, stream, di, cfg, bufs);
     return (int)hipGetLastError();
 }
 
