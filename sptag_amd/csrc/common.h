@@ -68,6 +68,8 @@ struct SearchBufs {
     void* gheap_spt;       /* nq * (spt_cap+1) * 8B */
 };
 
+enum { SPT_LDS_TIER_H = 255 };  /* SPT heap LDS tier entries (see kernels) */
+
 /* LDS bytes needed per workgroup for the LDS-heap variant (and the
  * non-heap part of the global variant). Keep in sync with kernel. */
 inline size_t lds_bytes(int dim, size_t esz, const SearchCfg& c, bool heaps_in_lds)
@@ -79,8 +81,9 @@ inline size_t lds_bytes(int dim, size_t esz, const SearchCfg& c, bool heaps_in_l
     b += ((size_t)c.k) * 8;                           /* result set */
     b += ((size_t)c.dpq_cap + 1) * 4;                 /* DistPriorityQueue */
     b += 64;                                          /* scalar slots, padding */
+    b += ((size_t)SPT_LDS_TIER_H + 1) * 8;            /* SPT heap LDS tier */
     if (heaps_in_lds) {
-        b += ((size_t)c.ng_cap + 1) * 8;   /* SPT heap is always global */
+        b += ((size_t)c.ng_cap + 1) * 8;   /* SPT heap tail is global */
     }
     return b;
 }

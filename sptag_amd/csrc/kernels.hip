@@ -48,11 +48,33 @@ struct QRes { int32_t vid; float dist; };
 /* 2^floor(log2(size)) — start of the heap's last level (Heap.h:25) */
 DEV int heap_lastlevel(int size) { return 1 << (31 - __clz(size)); }
 
+/* Binary min-heap, EXACT Heap.h semantics (pop order on equal keys is part
+ * of the parity contract), with TIERED storage: entries with index <=
+ * lsplit live in an LDS cache, the tail in `a` (global scratch or LDS).
+ * Storage placement never changes semantics — only access latency. The
+ * round-1 all-global SPT heap made the seed phase 41.5% of kernel cycles
+ * (profiles/r2 phase breakdown): every percolate level was an ~HBM-latency
+ * round trip. The hot top levels now stay in LDS. */
+enum { SPT_LDS_TIER = SPT_LDS_TIER_H };  /* 8 levels (1..255) + sentinel [0] */
+
 struct HeapRef {
-    NodeDist* a;   /* 1-based; a[0] = empty-top sentinel {-1, MAXDIST} */
-    int cap;       /* this buffer's capacity */
-    int ref_cap;   /* the reference's capacity for this heap */
+    NodeDist* a;    /* 1-based backing array; [0] = sentinel {-1, MAXDIST} */
+    NodeDist* lds;  /* optional LDS cache of [0..lsplit]; null = none */
+    int lsplit;
+    int cap;        /* this buffer's capacity */
+    int ref_cap;    /* the reference's capacity for this heap */
 };
+
+DEV NodeDist hget(const HeapRef& h, int i)
+{
+    return (h.lds && i <= h.lsplit) ? h.lds[i] : h.a[i];
+}
+
+DEV void hset(const HeapRef& h, int i, NodeDist v)
+{
+    if (h.lds && i <= h.lsplit) h.lds[i] = v;
+    else h.a[i] = v;
+}
 
 /* Heap.h:38-62 insert (incl. the full-heap last-level replace path). */
 DEV void ndheap_insert(HeapRef h, int* count, NodeDist v, int* oflow)
@@ -62,49 +84,70 @@ DEV void ndheap_insert(HeapRef h, int* count, NodeDist v, int* oflow)
         if (h.cap < h.ref_cap) { *oflow = 1; return; }
         int lastlevel = heap_lastlevel(h.cap);
         int maxi = lastlevel;
-        for (int i = lastlevel + 1; i <= h.cap; i++)
-            if (h.a[maxi].distance < h.a[i].distance) maxi = i;
-        if (v.distance > h.a[maxi].distance) return;
+        float maxd = hget(h, maxi).distance;
+        for (int i = lastlevel + 1; i <= h.cap; i++) {
+            float di = hget(h, i).distance;
+            if (maxd < di) { maxi = i; maxd = di; }
+        }
+        if (v.distance > maxd) return;
         loc = maxi;
     } else {
         loc = ++(*count);
     }
     int par = loc >> 1;
-    while (par > 0 && v.distance < h.a[par].distance) {
-        h.a[loc] = h.a[par];
+    while (par > 0) {
+        NodeDist p = hget(h, par);
+        if (!(v.distance < p.distance)) break;
+        hset(h, loc, p);
         loc = par;
         par >>= 1;
     }
-    h.a[loc] = v;
+    hset(h, loc, v);
 }
 
-/* Heap.h:90-105 heapify + :74-82 pop */
+/* Heap.h:90-105 heapify + :74-82 pop. The sinking element is the original
+ * root value throughout (the reference swaps it level by level), so it is
+ * carried in a register instead of re-read. */
 DEV void ndheap_heapify(HeapRef h, int count)
 {
     int parent = 1, next = 2;
+    if (next > count) return;
+    NodeDist pv = hget(h, parent);
     while (next < count) {
-        if (h.a[next].distance > h.a[next + 1].distance) next++;
-        if (h.a[next].distance < h.a[parent].distance) {
-            NodeDist t = h.a[parent]; h.a[parent] = h.a[next]; h.a[next] = t;
+        NodeDist nv = hget(h, next);
+        NodeDist nv2 = hget(h, next + 1);
+        if (nv.distance > nv2.distance) { next++; nv = nv2; }
+        if (nv.distance < pv.distance) {
+            hset(h, parent, nv);
+            hset(h, next, pv);
             parent = next;
             next <<= 1;
-        } else break;
+        } else {
+            return;
+        }
     }
-    if (next == count && h.a[next].distance < h.a[parent].distance) {
-        NodeDist t = h.a[parent]; h.a[parent] = h.a[next]; h.a[next] = t;
+    if (next == count) {
+        NodeDist nv = hget(h, next);
+        if (nv.distance < pv.distance) {
+            hset(h, parent, nv);
+            hset(h, next, pv);
+        }
     }
 }
 
 DEV NodeDist ndheap_pop(HeapRef h, int* count)
 {
-    if (*count == 0) return h.a[0];
-    NodeDist t = h.a[1]; h.a[1] = h.a[*count]; h.a[*count] = t;
+    if (*count == 0) return hget(h, 0);
+    NodeDist top = hget(h, 1);
+    hset(h, 1, hget(h, *count));
+    /* the vacated slot beyond count is dead until the next insert writes
+     * it (Heap.h swaps the old top there; nothing ever reads it) */
     (*count)--;
     ndheap_heapify(h, *count);
-    return h.a[*count + 1];
+    return top;
 }
 
-DEV NodeDist ndheap_top(HeapRef h, int count) { return count == 0 ? h.a[0] : h.a[1]; }
+DEV NodeDist ndheap_top(HeapRef h, int count) { return count == 0 ? hget(h, 0) : hget(h, 1); }
 
 /* DistPriorityQueue (WorkSpace.h:167-225): 1-based bounded float max-heap
  * seeded with MaxDist; len starts at 1. */
@@ -570,11 +613,15 @@ void bkt_search_kernel(DevIndex di, SearchCfg cfg, SearchBufs bufs)
 
     HeapRef ng, spt;
     ng.cap = cfg.ng_cap;  ng.ref_cap = cfg.max_check * 30;   /* WorkSpace.h:265 */
+    ng.lds = nullptr; ng.lsplit = 0;
     spt.cap = cfg.spt_cap; spt.ref_cap = cfg.max_check * 10;
-    /* SPT (tree) heap always lives in global scratch: it is touched only in
-     * the short seed/re-seed phases, and freeing its LDS doubles resident
-     * queries per CU. */
+    /* SPT (tree) heap: hot top levels in an LDS tier, tail in global
+     * scratch — full reference capacity without the LDS cost of the whole
+     * heap (its all-global round-1 form made the seed phase 41.5% of
+     * kernel cycles). */
     spt.a = (NodeDist*)bufs.gheap_spt + (size_t)q * (cfg.spt_cap + 1);
+    spt.lds = (NodeDist*)(smem + off); off += ((size_t)SPT_LDS_TIER + 1) * 8;
+    spt.lsplit = SPT_LDS_TIER;
     if (LDSHEAP) {
         ng.a = (NodeDist*)(smem + off); off += ((size_t)cfg.ng_cap + 1) * 8;
     } else {
@@ -589,8 +636,8 @@ void bkt_search_kernel(DevIndex di, SearchCfg cfg, SearchBufs bufs)
         ss->ng_count = 0; ss->spt_count = 0; ss->dpq_len = 1; ss->checked = 0;
         ss->oflow = 0; ss->terminate = 0; ss->break_flag = 0; ss->want_tree = 0;
         ng.a[0] = NodeDist{-1, MAXDIST};   /* Heap empty-top sentinel */
-        spt.a[0] = NodeDist{-1, MAXDIST};
-        dpq[1] = MAXDIST;                  /* DistPriorityQueue seed */
+        hset(spt, 0, NodeDist{-1, MAXDIST});
+        dpq[0] = MAXDIST;                  /* DistPriorityQueue seed element */
         for (int i = 0; i < cfg.k; i++) qrs[i] = QRes{-1, MAXDIST};
     }
     __syncthreads();
@@ -598,6 +645,13 @@ void bkt_search_kernel(DevIndex di, SearchCfg cfg, SearchBufs bufs)
     QCtx<T> c{&di, &cfg, qlds, dstage, istage, qrs, dpq, ng, spt,
               bufs.visited + (size_t)q * cfg.vcap, (uint32_t)(cfg.vcap - 1),
               ss, lane};
+
+    /* DistPriorityQueue (WorkSpace.h:167-225) as a FLAT bounded max-set:
+     * its observable behavior (the kept multiset and its current maximum)
+     * is independent of heap layout, so the m_Results state here is an
+     * unordered LDS array + uniform registers replicated wave-wide. */
+    float dmax = MAXDIST;
+    int dlen = 1, dargmax = 0;
 
     PMARK();
     init_search_trees_dev<T, DM>(c);
@@ -657,7 +711,7 @@ void bkt_search_kernel(DevIndex di, SearchCfg cfg, SearchBufs bufs)
                 }
             } else {
                 if (cfg.search_deleted || not_deleted(di, gnode.node)) {
-                    if (gnode.distance > dpq[1] || ss->checked > cfg.max_check)
+                    if (gnode.distance > dmax || ss->checked > cfg.max_check)
                         ss->terminate = 1;
                 }
             }
@@ -694,13 +748,41 @@ void bkt_search_kernel(DevIndex di, SearchCfg cfg, SearchBufs bufs)
         PACC(4);
         stage_dists<T, DM>(c, ncand);
         PACC(5);
-        if (lane == 0) {
-            for (int r = 0; r < ncand; r++) {
-                float dv = dstage[r];
-                ss->checked++;
-                if (dpq_insert(dpq, &ss->dpq_len, cfg.dpq_cap, dv))
-                    ndheap_insert(c.ng, &ss->ng_count, NodeDist{istage[r], dv}, &ss->oflow);
+        /* insert phase (BKTIndex.cpp:337-344). m_Results (the flat max-set)
+         * appends in O(1); a replace costs one wave-parallel max rescan
+         * (any max-valued slot may be replaced — same multiset). The NG
+         * frontier heap stays an exact Heap.h emulation on lane 0: its pop
+         * order on equal keys is part of the parity contract. */
+        for (int r = 0; r < ncand; r++) {
+            float dv = dstage[r];
+            if (lane == 0) ss->checked++;
+            if (!(dv > dmax)) {
+                if (dlen < cfg.dpq_cap) {
+                    if (lane == 0) dpq[dlen] = dv;
+                    dlen++;
+                } else {
+                    if (lane == 0) dpq[dargmax] = dv;
+                    __syncthreads();
+                    float m = -MAXDIST;
+                    int mi = 0;
+                    for (int i = lane; i < cfg.dpq_cap; i += 64) {
+                        float x = dpq[i];
+                        if (x > m) { m = x; mi = i; }
+                    }
+                    for (int o2 = 32; o2 > 0; o2 >>= 1) {
+                        float om = __shfl_down(m, o2);
+                        int omi = __shfl_down(mi, o2);
+                        if (om > m) { m = om; mi = omi; }
+                    }
+                    dmax = __shfl(m, 0);
+                    dargmax = __shfl(mi, 0);
+                }
+                if (lane == 0)
+                    ndheap_insert(c.ng, &ss->ng_count, NodeDist{istage[r], dv},
+                                  &ss->oflow);
             }
+        }
+        if (lane == 0) {
             /* dynamic pivots (BKTIndex.cpp:346-349) */
             ss->want_tree = (ndheap_top(c.ng, ss->ng_count).distance >
                              ndheap_top(c.spt, ss->spt_count).distance);
@@ -793,6 +875,9 @@ void bkt_iter_kernel(DevIndex di, SearchCfg cfg, IterBufs ib, int batch)
     HeapRef ng, spt;
     ng.cap = cfg.ng_cap;  ng.ref_cap = cfg.ng_cap;   /* reference capacities */
     spt.cap = cfg.spt_cap; spt.ref_cap = cfg.spt_cap;
+    /* untiered: heap state persists in global scratch across Next() calls */
+    ng.lds = nullptr; ng.lsplit = 0;
+    spt.lds = nullptr; spt.lsplit = 0;
     ng.a = (NodeDist*)ib.gheap_ng + (size_t)q * (cfg.ng_cap + 1);
     spt.a = (NodeDist*)ib.gheap_spt + (size_t)q * (cfg.spt_cap + 1);
 
@@ -1029,8 +1114,11 @@ void kdt_search_kernel(DevIndex di, SearchCfg cfg, SearchBufs bufs)
 
     HeapRef ng, spt;
     ng.cap = cfg.ng_cap;  ng.ref_cap = cfg.max_check * 30;
+    ng.lds = nullptr; ng.lsplit = 0;
     spt.cap = cfg.spt_cap; spt.ref_cap = cfg.max_check * 10;
     spt.a = (NodeDist*)bufs.gheap_spt + (size_t)q * (cfg.spt_cap + 1);
+    spt.lds = (NodeDist*)(smem + off); off += ((size_t)SPT_LDS_TIER + 1) * 8;
+    spt.lsplit = SPT_LDS_TIER;
     if (LDSHEAP) {
         ng.a = (NodeDist*)(smem + off); off += ((size_t)cfg.ng_cap + 1) * 8;
     } else {
@@ -1045,7 +1133,7 @@ void kdt_search_kernel(DevIndex di, SearchCfg cfg, SearchBufs bufs)
         ss->oflow = 0; ss->terminate = 0; ss->break_flag = 0; ss->want_tree = 0;
         ss->tree_checked = 0; ss->no_better = 0;
         ng.a[0] = NodeDist{-1, MAXDIST};
-        spt.a[0] = NodeDist{-1, MAXDIST};
+        hset(spt, 0, NodeDist{-1, MAXDIST});
         dpq[1] = MAXDIST;
         for (int i = 0; i < cfg.k; i++) qrs[i] = QRes{-1, MAXDIST};
     }
