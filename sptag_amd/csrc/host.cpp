@@ -21,6 +21,7 @@
 
 #include <cstdio>
 #include <cstdlib>
+#include <algorithm>
 #include <cstring>
 #include <mutex>
 #include <string>
@@ -677,7 +678,10 @@ SptagAmdIterBatch* sptag_amd_iter_create(SptagAmdIndex* ix, const void* queries,
 {
     if (!ix || !queries || nq <= 0) return nullptr;
     if (ix->algo != ALGO_BKT) {
-        fprintf(stderr, "sptag_amd: iterative search is BKT-only\n");
+        /* reference parity: KDT::Index<T>::GetIterator logs "ITERATIVE NOT
+         * SUPPORT FOR KDT" and returns null (KDTIndex.cpp:322-346) */
+        fprintf(stderr, "sptag_amd: iterative search is BKT-only "
+                        "(as in the reference: KDTIndex.cpp:322)\n");
         return nullptr;
     }
     if (!sptag_amd_gpu_available() || !ix->d_vectors) {
@@ -1011,9 +1015,21 @@ int sptag_amd_add(SptagAmdIndex* ix, const void* vectors, int32_t nadd,
     int32_t* d_v = nullptr;
     float* d_d = nullptr;
     void* d_q = nullptr;
+    /* staging for the batched changed-row upload: one H2D + one scatter
+     * per added node instead of one 128 B hipMemcpy per touched edge (the
+     * round-1 form did up to ~500 tiny copies per add). The SEARCH stays
+     * sequential per node — reference AddIndex semantics (BKTIndex.cpp:
+     * 966-969: each refine sees every previously added node's edges). */
+    int32_t* d_rows = nullptr;     /* staged row payloads */
+    int32_t* d_rowidx = nullptr;   /* their row ids */
     HIP_OR_FAIL(hipMalloc(&d_v, (size_t)k * 4), SPTAG_AMD_ERR_OOM);
     HIP_OR_FAIL(hipMalloc(&d_d, (size_t)k * 4), SPTAG_AMD_ERR_OOM);
     HIP_OR_FAIL(hipMalloc(&d_q, (size_t)ix->dim * esz), SPTAG_AMD_ERR_OOM);
+    HIP_OR_FAIL(hipMalloc(&d_rows, (size_t)(k + 1) * ix->deg * 4),
+                SPTAG_AMD_ERR_OOM);
+    HIP_OR_FAIL(hipMalloc(&d_rowidx, (size_t)(k + 1) * 4), SPTAG_AMD_ERR_OOM);
+    std::vector<int32_t> hrows((size_t)(k + 1) * ix->deg);
+    std::vector<int32_t> hidx(k + 1);
     int rc = SPTAG_AMD_OK;
     for (int32_t node = begin; node < end && rc == SPTAG_AMD_OK; node++) {
         HIP_OR_FAIL(hipMemcpy(d_q, hvec(ix, node), (size_t)ix->dim * esz,
@@ -1028,24 +1044,36 @@ int sptag_amd_add(SptagAmdIndex* ix, const void* vectors, int32_t nadd,
         HIP_OR_FAIL(hipMemcpy(rd.data(), d_d, (size_t)k * 4, hipMemcpyDeviceToHost),
                     SPTAG_AMD_ERR_NOGPU);
         host_rebuild_neighbors(ix, node, rv.data(), rd.data(), k);
-        /* upload the refined row, then the two-way inserts */
-        HIP_OR_FAIL(hipMemcpy(ix->d_graph + (size_t)node * ix->deg,
-                              ix->h_graph.data() + (size_t)node * ix->deg,
-                              (size_t)ix->deg * 4, hipMemcpyHostToDevice),
-                    SPTAG_AMD_ERR_NOGPU);
+        int nrows = 0;
+        hidx[nrows++] = node;
         for (int j = 0; j < k; j++) {
             if (rv[j] < 0) break;
             if (rv[j] == node) continue;
             host_insert_neighbors(ix, rv[j], node, rd[j]);
-            HIP_OR_FAIL(hipMemcpy(ix->d_graph + (size_t)rv[j] * ix->deg,
-                                  ix->h_graph.data() + (size_t)rv[j] * ix->deg,
-                                  (size_t)ix->deg * 4, hipMemcpyHostToDevice),
-                        SPTAG_AMD_ERR_NOGPU);
+            hidx[nrows++] = rv[j];
         }
+        /* dedup row ids, stage payloads, one H2D + one scatter */
+        std::sort(hidx.begin(), hidx.begin() + nrows);
+        nrows = (int)(std::unique(hidx.begin(), hidx.begin() + nrows) -
+                      hidx.begin());
+        for (int j = 0; j < nrows; j++)
+            memcpy(&hrows[(size_t)j * ix->deg],
+                   ix->h_graph.data() + (size_t)hidx[j] * ix->deg,
+                   (size_t)ix->deg * 4);
+        HIP_OR_FAIL(hipMemcpy(d_rows, hrows.data(),
+                              (size_t)nrows * ix->deg * 4,
+                              hipMemcpyHostToDevice), SPTAG_AMD_ERR_NOGPU);
+        HIP_OR_FAIL(hipMemcpy(d_rowidx, hidx.data(), (size_t)nrows * 4,
+                              hipMemcpyHostToDevice), SPTAG_AMD_ERR_NOGPU);
+        launch_scatter_rows(ix->d_graph, d_rows, ix->deg * 4, d_rowidx, nrows,
+                            nullptr);
+        HIP_OR_FAIL(hipDeviceSynchronize(), SPTAG_AMD_ERR_NOGPU);
     }
     (void)hipFree(d_v);
     (void)hipFree(d_d);
     (void)hipFree(d_q);
+    (void)hipFree(d_rows);
+    (void)hipFree(d_rowidx);
     return rc;
 }
 

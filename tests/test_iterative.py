@@ -10,8 +10,10 @@ import os
 import numpy as np
 import pytest
 
-from conftest import GOLDEN, load_golden
+from conftest import GOLDEN, golden_fixtures, load_golden
 from oracle.pyoracle import OrcIndex
+
+import sptag_amd
 
 FIXTURES = ["f32_l2_n10k_d32", "i8_l2_n10k_d100"]
 
@@ -61,3 +63,18 @@ def test_gpu_iterative_bit_exact(name):
             np.testing.assert_array_equal(
                 dists[i, :cnt], rec[i, c, 3::2].view(np.float32)[:cnt])
     it.Close()
+
+
+@pytest.mark.gpu
+def test_kdt_iterate_refused_like_reference():
+    """Reference parity for the refusal itself: KDT::Index<T>::GetIterator
+    logs 'ITERATIVE NOT SUPPORT FOR KDT' and returns null
+    (src/Core/KDT/KDTIndex.cpp:322-346) — the backend's iterator create
+    rejects KDT handles the same way."""
+    names = [n for n in golden_fixtures() if n.startswith("kdt")]
+    if not names:
+        pytest.skip("no KDT fixtures")
+    g = load_golden(names[0])
+    ix = sptag_amd.AnnIndex.Load(g["index"])
+    with pytest.raises(Exception):
+        ix.Iterate(g["queries"][:4], max_check=512)
