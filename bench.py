@@ -288,6 +288,10 @@ def main():
     ap.add_argument("--srefine-k", type=int, default=0)
     ap.add_argument("--srefine-mc", type=int, default=0)
     ap.add_argument("--fill-pruned", type=int, default=-1)
+    ap.add_argument("--force-shard", default="",
+                    help="R/W: build+search shard R of W on ONE process "
+                         "(no collectives; shard-local recall) — verifies "
+                         "the config-#5 per-shard path on a single GPU")
     args = ap.parse_args()
 
     import torch
@@ -297,8 +301,13 @@ def main():
     world = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
     local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    forced_shard = False
+    if args.force_shard:
+        r, w = args.force_shard.split("/")
+        rank, world, local_rank = int(r), int(w), 0
+        forced_shard = True
     dist = None
-    if world > 1:
+    if world > 1 and not forced_shard:
         import torch.distributed as tdist
         dist = tdist
         dist.init_process_group("nccl")
@@ -353,7 +362,7 @@ def main():
     torch.cuda.empty_cache()
     t0 = time.time()
     tv, td = shard_truth(xs, q, cfg["k"], cfg["metric"], torch, lo)
-    if world > 1:
+    if dist:
         gv = [torch.zeros_like(tv) for _ in range(world)]
         gd = [torch.zeros_like(td) for _ in range(world)]
         dist.all_gather(gv, tv.contiguous())
@@ -382,7 +391,7 @@ def main():
         world > 1."""
         one_step(mc)
         v = (d_vids + lo).masked_fill(d_vids < 0, -1)
-        if world == 1:
+        if not dist:
             return v, d_dists
         gv = [torch.empty_like(v) for _ in range(world)]
         gd = [torch.empty_like(d_dists) for _ in range(world)]
@@ -412,17 +421,17 @@ def main():
     # all-gather + device merge
     for _ in range(args.warmup):
         step_merged(chosen_mc)
-    if world > 1:
+    if dist:
         dist.barrier()
     torch.cuda.synchronize()
     t0 = time.time()
     for _ in range(args.steps):
         step_merged(chosen_mc)
     torch.cuda.synchronize()
-    if world > 1:
+    if dist:
         dist.barrier()
     elapsed = time.time() - t0
-    if world > 1:
+    if dist:
         te = torch.tensor([elapsed], device=device)
         dist.all_reduce(te, op=dist.ReduceOp.MAX)
         elapsed = float(te.item())
@@ -454,9 +463,9 @@ def main():
     ms_per_step = elapsed / args.steps * 1e3
 
     result = None
-    if rank == 0:
+    if rank == 0 or forced_shard:
         cpu = None
-        if not args.no_cpu_baseline and world == 1:
+        if not args.no_cpu_baseline and not dist and not forced_shard:
             t0 = time.time()
             idx_dir = os.environ.get("BENCH_INDEX_DIR", "/tmp/bench_index")
             os.makedirs(idx_dir, exist_ok=True)
