@@ -105,3 +105,29 @@ def test_kdt_tree_routes_every_point():
             stack.append(int(nodes[nd, 0]))
             stack.append(int(nodes[nd, 1]))
         assert len(seen) == n, (t, len(seen))
+
+
+def test_rng_prune_fill_pruned():
+    """fill_pruned pads leftover degree slots with the nearest REJECTED
+    candidates (billion-scale build knob); the RNG-accepted prefix must be
+    identical to the strict prune, and the filled row stays ascending."""
+    torch.manual_seed(5)
+    n, c, d, deg = 40, 24, 8, 12
+    x = torch.randn(n, d)
+    cdd0 = torch.rand(n, c).sort(dim=1).values * 10
+    cid = torch.stack([torch.randperm(n)[:c] for _ in range(n)]).int()
+    strict = _rng_prune(x, cid, cdd0, deg, 1.0, "cpu").numpy()
+    filled = _rng_prune(x, cid, cdd0, deg, 1.0, "cpu",
+                        fill_pruned=True).numpy()
+    for i in range(n):
+        s = strict[i][strict[i] >= 0]
+        f = filled[i][filled[i] >= 0]
+        assert len(f) >= len(s)
+        assert len(f) == min(deg, c)  # all slots used (valid pool is full)
+        # accepted set preserved (as a subsequence of the filled row)
+        it = iter(f.tolist())
+        assert all(v in it for v in s.tolist()) or set(s) <= set(f)
+        # filled row ascending by the pool's distance order
+        pos = {int(v): j for j, v in enumerate(cid[i].tolist())}
+        order = [pos[int(v)] for v in f.tolist()]
+        assert order == sorted(order)
