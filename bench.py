@@ -59,20 +59,23 @@ CONFIGS = {
                                  metric="Cosine", nq=10_000, k=10,
                                  ncenters=16384, sigma=30.0, cand=128,
                                  ntrees=4, refine=2, srefine=1,
-                                 srefine_k=256, srefine_mc=2048),
+                                 srefine_k=512, srefine_mc=2048,
+                                 fill_pruned=True),
     # 30M validation scale for the config-#3 build recipe
     "bkt_30m_d100_i8_cos": dict(n=30_000_000, d=100, dtype="i8",
                                 metric="Cosine", nq=10_000, k=10,
                                 ncenters=16384, sigma=30.0, cand=128,
                                 ntrees=4, refine=2, srefine=1,
-                                srefine_k=256, srefine_mc=2048),
+                                srefine_k=512, srefine_mc=2048,
+                                fill_pruned=True),
     # BASELINE.json configs[4] — 1B int8 L2, meant for --gpus 8 (125M rows
     # per shard; per-shard recipe = the config-#3 recipe)
     "bkt_1b_d100_i8_l2": dict(n=1_000_000_000, d=100, dtype="i8",
                               metric="L2", nq=10_000, k=10,
                               ncenters=65536, sigma=30.0, cand=128,
                               ntrees=4, refine=2, srefine=1,
-                              srefine_k=256, srefine_mc=2048),
+                              srefine_k=512, srefine_mc=2048,
+                              fill_pruned=True),
     # BASELINE.json configs[3] — KDT, embedding shape. Note: the KDT
     # algorithm's no-better-propagation termination caps recall on this
     # data family at ~0.93-0.94 for the REFERENCE implementation as well
@@ -120,6 +123,19 @@ def gen_data(cfg, shard, world, device, torch):
         # embedding-shaped: gaussian centers scaled vs unit noise (config #4)
         centers = torch.randn((cfg["ncenters"], d), generator=gen,
                               device=device) * cfg["sigma"]
+    elif cfg["dtype"] == "i8":
+        # SPACEV-shaped (int8 values ~U[-100,100]): the
+        # hierarchical mixture centered at 0 so the int8 clamp at +-127
+        # only cuts a rare tail (7% of coords). Round 1 reused the SIFT
+        # [0,255] box here; the clamp then saturated 50% of ALL
+        # coordinates at exactly +127 — half the dimensions binarized, a
+        # data-generator bug that degrades metric structure at scale.
+        # Round-1 int8 lines were measured on that flawed data.
+        supers = torch.rand((256, d), generator=gen, device=device) * 160.0 - 80.0
+        slab = torch.randint(0, 256, (cfg["ncenters"],), generator=gen,
+                             device=device)
+        centers = supers[slab] + torch.randn((cfg["ncenters"], d), generator=gen,
+                                             device=device) * (cfg["sigma"] * 1.5)
     else:
         # SIFT-shaped: hierarchical mixture over the [0,255] box (super-
         # centers -> centers -> points; BASELINE.md clusterability note)

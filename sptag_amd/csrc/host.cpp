@@ -503,6 +503,11 @@ static int search_device_core(SptagAmdIndex* ix, const void* d_q, int32_t nq,
             for (int p = 0; p < 10; p++)
                 fprintf(stderr, "  %-7s %14.0f  (%5.1f%%)\n", pname[p], tot[p],
                         all > 0 ? 100.0 * tot[p] / all : 0.0);
+            double ngi = 0;
+            for (int32_t i = 0; i < nq; i++)
+                ngi += (double)stats[(size_t)sstride * i + 12];
+            fprintf(stderr, "  (insert = ng-heap %14.0f + results-set rest)\n",
+                    ngi);
         }
         if (!redo.empty() && getenv("SPTAG_AMD_DEBUG"))
             fprintf(stderr, "sptag_amd: %zu/%d queries overflowed (rerun)\n",

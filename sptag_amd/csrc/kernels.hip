@@ -497,6 +497,28 @@ DEV void stage_dists(QCtx<T>& c, int cnt)
     __syncthreads();
 }
 
+/* Warm the GLOBAL parent-chain levels of the next `cnt` SPT insertion
+ * slots (slot for the i-th insert is spt_count+1+i while the heap is not
+ * full — seed-phase heaps never reach ref_cap in practice, and the warm is
+ * advisory: a wrong slot guess only wastes a touch). Lane-0's serial
+ * percolate-up then reads L1/L2-warm lines instead of paying an HBM
+ * round trip per level — the seed phase was 31% of kernel cycles with
+ * cold parent reads. */
+template <typename T>
+DEV void spt_warm_parents(QCtx<T>& c, int cnt)
+{
+    int loc = c.ss->spt_count + 1 + c.lane;
+    uint32_t wsink = 0;
+    if (c.lane < cnt) {
+        int par = loc >> 1;
+        while (par > c.spt.lsplit) {
+            wsink += ((const uint8_t*)&c.spt.a[par])[0];
+            par >>= 1;
+        }
+    }
+    asm volatile("" :: "v"(wsink));
+}
+
 /* BKTree.h:772 SearchTrees — wave-cooperative, serial decisions on lane 0. */
 template <typename T, int DM>
 DEV void search_trees_dev(QCtx<T>& c, int limit)
@@ -532,6 +554,7 @@ DEV void search_trees_dev(QCtx<T>& c, int limit)
                 if (c.lane < cnt) c.istage[c.lane] = di.tree_nodes[(size_t)(base + c.lane) * 3];
                 __syncthreads();
                 stage_dists<T, DM>(c, cnt);
+                spt_warm_parents(c, cnt);
                 if (c.lane == 0)
                     for (int i = 0; i < cnt; i++)
                         ndheap_insert(c.spt, &c.ss->spt_count,
@@ -566,6 +589,7 @@ DEV void init_search_trees_dev(QCtx<T>& c)
                 if (c.lane < cnt) c.istage[c.lane] = di.tree_nodes[(size_t)(base + c.lane) * 3];
                 __syncthreads();
                 stage_dists<T, DM>(c, cnt);
+                spt_warm_parents(c, cnt);
                 if (c.lane == 0)
                     for (int i = 0; i < cnt; i++)
                         ndheap_insert(c.spt, &c.ss->spt_count,
@@ -599,7 +623,7 @@ void bkt_search_kernel(DevIndex di, SearchCfg cfg, SearchBufs bufs)
     if (q >= cfg.nq) return;
     const int lane = threadIdx.x;
     uint64_t pc[10] = {};
-    uint64_t pmark = 0;
+    uint64_t pmark = 0, pc_ng = 0;
 
     extern __shared__ char smem[];
     size_t off = 0;
@@ -777,9 +801,17 @@ void bkt_search_kernel(DevIndex di, SearchCfg cfg, SearchBufs bufs)
                     dmax = __shfl(m, 0);
                     dargmax = __shfl(mi, 0);
                 }
-                if (lane == 0)
-                    ndheap_insert(c.ng, &ss->ng_count, NodeDist{istage[r], dv},
-                                  &ss->oflow);
+                if (lane == 0) {
+                    if (PROF) {
+                        uint64_t t0_ = clock64();
+                        ndheap_insert(c.ng, &ss->ng_count,
+                                      NodeDist{istage[r], dv}, &ss->oflow);
+                        pc_ng += clock64() - t0_;
+                    } else {
+                        ndheap_insert(c.ng, &ss->ng_count,
+                                      NodeDist{istage[r], dv}, &ss->oflow);
+                    }
+                }
             }
         }
         if (lane == 0) {
@@ -831,7 +863,8 @@ void bkt_search_kernel(DevIndex di, SearchCfg cfg, SearchBufs bufs)
                 s[1] = popped;
 #pragma unroll
                 for (int i = 0; i < 10; i++) s[2 + i] = (int32_t)pc[i];
-                s[12] = s[13] = s[14] = s[15] = 0;
+                s[12] = (int32_t)pc_ng;   /* NG-insert share of `insert` */
+                s[13] = s[14] = s[15] = 0;
             } else {
                 bufs.stats[(size_t)q * 2 + 0] = ss->checked;
                 bufs.stats[(size_t)q * 2 + 1] = popped;
