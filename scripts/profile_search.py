@@ -19,7 +19,16 @@ import sptag_amd  # noqa: E402
 def main():
     idx_dir, qfile, mc, warmup, steps = sys.argv[1:6]
     mc, warmup, steps = int(mc), int(warmup), int(steps)
-    q = np.load(qfile)
+    if qfile.endswith(".npy"):
+        q = np.load(qfile)
+    else:   # reference DEFAULT binary format [int32 n][int32 d][rows]
+        with open(qfile, "rb") as f:
+            n, d = np.frombuffer(f.read(8), dtype=np.int32)
+            import sptag_amd as _s
+            ix0 = _s.AnnIndex.Load(idx_dir)
+            dt = np.float32 if ix0.valuetype == 0 else np.int8
+            del ix0
+            q = np.frombuffer(f.read(), dtype=dt).reshape(n, d)
     ix = sptag_amd.AnnIndex.Load(idx_dir)
     k = 10
     for _ in range(warmup):
