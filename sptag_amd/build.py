@@ -719,12 +719,17 @@ def refine_via_search(vectors_t, tree_start, tree_nodes, graph, cand_ids,
     xf = vectors_t if vectors_t.dtype == torch.float32 else vectors_t.float()
     self_ids = torch.arange(n, device=device, dtype=torch.int32)
     if chunk <= 0:
-        # the searcher's per-query visited table is 4*next_pow2(4*max_check)
-        # bytes; keep the whole-batch scratch ~8 GB
+        # scratch-bounded chunk: visited table (scales with mc AND k — deep
+        # results queues keep the traversal expanding past MaxCheck) at
+        # ~8 GB, plus headroom for the overflow rerun's reference-capacity
+        # global heaps (40*mc entries x 8 B per query) at ~12 GB.
         vcap = 1
-        while vcap < max(4096, max_check * 4):
+        while vcap < max(4096, max_check * 4 + 64 * k):
             vcap <<= 1
-        chunk = min(1_000_000, max(65_536, int(8e9 // (vcap * 4))))
+        chunk = min(1_000_000,
+                    int(8e9 // (vcap * 4)),
+                    int(12e9 // (max_check * 40 * 8)))
+        chunk = max(16_384, chunk)
     x_np = vectors_t.cpu().numpy()
     graph_t = torch.as_tensor(graph, device=device)         if not torch.is_tensor(graph) else graph
     # Search distances arrive in the index metric; the builder's pools are

@@ -406,7 +406,13 @@ static int search_device_core(SptagAmdIndex* ix, const void* d_q, int32_t nq,
     cfg.search_deleted = 0;
     if (ix->refine_mode) { cfg.search_dup = 0; cfg.search_deleted = 1; }
     cfg.dpq_cap = std::max(max_check / 16, k);   /* WorkSpace.h:268 */
-    cfg.vcap = (int32_t)next_pow2((uint32_t)std::max(4096, max_check * 4));
+    /* k-deep (refine-class) searches keep expanding far past MaxCheck —
+     * the budget is only consulted when a pop misses the k-deep results
+     * queue (BKTIndex.cpp:326) — so the visited table must scale with k
+     * as well (the reference's OptHashPosVector grows on demand,
+     * WorkSpace.h:119; a fixed GPU table errors loudly on saturation). */
+    cfg.vcap = (int32_t)next_pow2(
+        (uint32_t)std::max(4096, max_check * 4 + 64 * k));
     /* LDS-variant capacities: sized to the TYPICAL traversal occupancy so
      * several workgroups fit per CU; the rare query that outgrows them is
      * rerun on the global-heap variant at the reference's own capacities
@@ -414,6 +420,7 @@ static int search_device_core(SptagAmdIndex* ix, const void* d_q, int32_t nq,
     /* KDT inserts every computed neighbor into the frontier (no results-
      * queue gate), so its occupancy tracks `checked` itself. */
     cfg.ng_cap = ix->algo == ALGO_KDT ? max_check + 2048 : max_check / 2 + 512;
+    if (k > 64) cfg.ng_cap += 4 * k;   /* k-deep frontiers run larger */
     if (const char* e = getenv("SPTAG_AMD_NG_CAP"))   /* perf experiments */
         cfg.ng_cap = std::max(256, atoi(e));
     cfg.spt_cap = 4096;
