@@ -304,6 +304,8 @@ def main():
     ap.add_argument("--srefine-k", type=int, default=0)
     ap.add_argument("--srefine-mc", type=int, default=0)
     ap.add_argument("--fill-pruned", type=int, default=-1)
+    ap.add_argument("--init-pivots", type=int, default=0,
+                    help="NumberOfInitialDynamicPivots override (seed leaves)")
     ap.add_argument("--force-shard", default="",
                     help="R/W: build+search shard R of W on ONE process "
                          "(no collectives; shard-local recall) — verifies "
@@ -366,6 +368,9 @@ def main():
         ix = sptag_amd.AnnIndex.FromArrays(
             arrays["vectors"], arrays["tree_start"], arrays["tree_nodes"],
             arrays["graph"], cfg["metric"], device=local_rank)
+    ipv = args.init_pivots or cfg.get("init_pivots", 0)
+    if ipv:
+        ix.SetSearchParams(init_pivots=ipv)
     log(rank, f"index uploaded ({time.time()-t0:.1f}s)")
 
     # truth runs in the STORED vector space (cosine bases are normalized on

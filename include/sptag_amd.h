@@ -154,6 +154,16 @@ int64_t sptag_amd_deleted_count(const SptagAmdIndex* idx);
  * (vectors/tree/graph/deletes + indexloader.ini). */
 int sptag_amd_save_index(SptagAmdIndex* idx, const char* folder);
 
+/* Search-parameter setter — the C-ABI form of the reference's
+ * SetParameter("NumberOfInitialDynamicPivots"/"NumberOfOtherDynamicPivots"/
+ * "ThresholdOfNumberOfContinuousNoBetterPropagation"/"MaxCheck", ...)
+ * (BKTIndex.cpp:980, ParameterDefinitionList.h). Pass <=0 to keep a
+ * value unchanged. */
+void sptag_amd_set_search_params(SptagAmdIndex* idx, int32_t init_pivots,
+                                 int32_t other_pivots,
+                                 int32_t nobetter_threshold,
+                                 int32_t default_maxcheck);
+
 /* metadata */
 int32_t sptag_amd_num_vectors(const SptagAmdIndex* idx);
 int32_t sptag_amd_dim(const SptagAmdIndex* idx);

@@ -97,6 +97,10 @@ def load_library():
                                         ctypes.c_void_p, ctypes.c_void_p,
                                         ctypes.c_void_p, ctypes.c_void_p]
     lib.sptag_amd_iter_free.argtypes = [ctypes.c_void_p]
+    lib.sptag_amd_set_search_params.restype = None
+    lib.sptag_amd_set_search_params.argtypes = [
+        ctypes.c_void_p, ctypes.c_int32, ctypes.c_int32, ctypes.c_int32,
+        ctypes.c_int32]
     lib.sptag_amd_gpu_available.restype = ctypes.c_int
     lib.sptag_amd_build_info.restype = ctypes.c_char_p
     _LIB = lib
@@ -328,6 +332,13 @@ class AnnIndex:
     @property
     def deleted_count(self):
         return self._lib.sptag_amd_deleted_count(self._h)
+
+    def SetSearchParams(self, init_pivots=0, other_pivots=0,
+                        nobetter_threshold=0, max_check=0):
+        """Reference SetParameter equivalents (NumberOfInitialDynamicPivots
+        etc., BKTIndex.cpp:980); pass 0 to keep a value."""
+        self._lib.sptag_amd_set_search_params(
+            self._h, init_pivots, other_pivots, nobetter_threshold, max_check)
 
     def Iterate(self, queries, max_check=0):
         """Streaming search: mirrors the reference GetIterator/Next protocol

@@ -1246,4 +1246,16 @@ int32_t sptag_amd_default_maxcheck(const SptagAmdIndex* ix)
 { return ix ? ix->default_maxcheck : -1; }
 int sptag_amd_algo(const SptagAmdIndex* ix) { return ix ? ix->algo : -1; }
 
+void sptag_amd_set_search_params(SptagAmdIndex* ix, int32_t init_pivots,
+                                 int32_t other_pivots,
+                                 int32_t nobetter_threshold,
+                                 int32_t default_maxcheck)
+{
+    if (!ix) return;
+    if (init_pivots > 0) ix->init_pivots = init_pivots;
+    if (other_pivots > 0) ix->other_pivots = other_pivots;
+    if (nobetter_threshold > 0) ix->nobetter_threshold = nobetter_threshold;
+    if (default_maxcheck > 0) ix->default_maxcheck = default_maxcheck;
+}
+
 }  /* extern "C" */
