@@ -47,14 +47,12 @@ CONFIGS = {
     # quick smoke-scale variant for plumbing runs (not a bench line)
     "bkt_1m_d128_f32_l2": dict(n=1_000_000, d=128, dtype="f32", metric="L2",
                                nq=10_000, k=10, ncenters=4096, sigma=32.0),
-    # BASELINE.json configs[2] — int8 cosine SPACEV shape
-    # NOTE round-1 status: the SEARCH kernel is bit-identical to the
-    # reference at this scale (scripts/diag_scale.py: GPU==oracle, 1.000),
-    # but the torch builder's candidate pools thin out at 100M (recall
-    # 0.26 @ mc16k with 2 trees; 0.93 at 30M with 10 trees + 2 refine
-    # rounds). Search-based refinement (the reference's own recipe,
-    # SURVEY.md §8f rank 1) is the round-2 fix; until then this config
-    # reports with recall_gate_met=false.
+    # BASELINE.json configs[2] — int8 cosine SPACEV shape. The search
+    # kernel is bit-identical to the reference at every scale
+    # (scripts/diag_scale.py); graph quality at 30M+ comes from the
+    # round-2 recipe: NN-descent rounds + wide-prune search-refine
+    # (reference RefineGraph semantics) + fill-pruned slots. See DESIGN.md
+    # §5 for the measured progression.
     "bkt_100m_d100_i8_cos": dict(n=100_000_000, d=100, dtype="i8",
                                  metric="Cosine", nq=10_000, k=10,
                                  ncenters=16384, sigma=30.0, cand=128,
