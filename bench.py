@@ -293,6 +293,8 @@ def main():
     ap.add_argument("--mc", type=int, default=0, help="force MaxCheck (skip sweep)")
     ap.add_argument("--no-cpu-baseline", action="store_true")
     ap.add_argument("--ntrees", type=int, default=4)
+    ap.add_argument("--degree", type=int, default=0,
+                    help="graph NeighborhoodSize (reference default 32)")
     ap.add_argument("--cand", type=int, default=0)
     ap.add_argument("--ncenters", type=int, default=0)
     ap.add_argument("--refine", type=int, default=-1,
@@ -343,6 +345,7 @@ def main():
     x_np = x.cpu().numpy()
     arrays = build_index_arrays(
         x_np, cfg["metric"], algo=cfg.get("algo", "BKT"),
+        degree=args.degree or cfg.get("degree", 32),
         cand=args.cand or cfg.get("cand", 256),
         kdt_trees=cfg.get("kdt_trees", 2),
         ntrees=cfg.get("ntrees", args.ntrees),
