@@ -24,7 +24,13 @@ import sptag_amd  # noqa: E402
 def main():
     idx_dir, nq = sys.argv[1], int(sys.argv[2])
     ix = sptag_amd.AnnIndex.Load(idx_dir)
-    q = np.load(os.path.join(idx_dir, "queries.npy"))[:nq]
+    rng = np.random.default_rng(7)
+    # query content is irrelevant to the byte count (the read volume is
+    # the streamed vector blob)
+    if ix.valuetype == 0:
+        q = rng.standard_normal((nq, ix.dim)).astype(np.float32)
+    else:
+        q = rng.integers(-100, 101, (nq, ix.dim)).astype(np.int8)
     v, d = ix.Truth(q, 10)
     esz = 4 if ix.valuetype == 0 else 1
     alg = nq * ix.n * ix.dim * esz
