@@ -55,23 +55,23 @@ CONFIGS = {
     # §5 for the measured progression.
     "bkt_100m_d100_i8_cos": dict(n=100_000_000, d=100, dtype="i8",
                                  metric="Cosine", nq=10_000, k=10,
-                                 ncenters=16384, sigma=30.0, cand=128,
-                                 ntrees=4, refine=2, srefine=1,
+                                 ncenters=16384, sigma=30.0, cand=160,
+                                 degree=48, ntrees=4, refine=2, srefine=1,
                                  srefine_k=512, srefine_mc=2048,
                                  fill_pruned=True),
     # 30M validation scale for the config-#3 build recipe
     "bkt_30m_d100_i8_cos": dict(n=30_000_000, d=100, dtype="i8",
                                 metric="Cosine", nq=10_000, k=10,
-                                ncenters=16384, sigma=30.0, cand=128,
-                                ntrees=4, refine=2, srefine=1,
+                                ncenters=16384, sigma=30.0, cand=160,
+                                degree=48, ntrees=4, refine=2, srefine=1,
                                 srefine_k=512, srefine_mc=2048,
                                 fill_pruned=True),
     # BASELINE.json configs[4] — 1B int8 L2, meant for --gpus 8 (125M rows
     # per shard; per-shard recipe = the config-#3 recipe)
     "bkt_1b_d100_i8_l2": dict(n=1_000_000_000, d=100, dtype="i8",
                               metric="L2", nq=10_000, k=10,
-                              ncenters=65536, sigma=30.0, cand=128,
-                              ntrees=4, refine=2, srefine=1,
+                              ncenters=65536, sigma=30.0, cand=160,
+                              degree=48, ntrees=4, refine=2, srefine=1,
                               srefine_k=512, srefine_mc=2048,
                               fill_pruned=True),
     # BASELINE.json configs[3] — KDT, embedding shape. Note: the KDT
