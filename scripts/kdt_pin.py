@@ -30,9 +30,9 @@ from sptag_amd.build import build_index_arrays  # noqa: E402
 SWEEP = [2048, 4096, 8192, 16384, 32768]
 
 
-def recall_curve(ix, q, tv, k):
+def recall_curve(ix, q, tv, k, sweep=None):
     out = {}
-    for mc in SWEEP:
+    for mc in (sweep or SWEEP):
         v, _ = ix.BatchSearch(q.cpu().numpy(), k, mc)
         out[mc] = bench.recall_at_k(v, tv, k)
     return out
@@ -78,14 +78,14 @@ def part_b():
     xs = torch.as_tensor(arrays["vectors"], device="cuda:0")
     tv, _ = bench.shard_truth(xs, q, cfg["k"], "Cosine", torch, 0)
     tv = tv.cpu().numpy()
-    cur = recall_curve(ix, q, tv, cfg["k"])
+    cur = recall_curve(ix, q, tv, cfg["k"], sweep=(4096, 8192, 16384))
     print("[kdt_pin B] GPU recall@10: " +
           " ".join(f"mc{m}={r:.4f}" for m, r in cur.items()), flush=True)
 
     idx = "/tmp/kdtpin_idx"
     os.makedirs(idx, exist_ok=True)
     ix.Save(idx)
-    nq = 2000   # bounded sample for the CPU reference
+    nq = 1500   # bounded sample for the CPU reference
     qs = q[:nq].cpu().numpy()
     qfile = os.path.join(idx, "q.bin")
     with open(qfile, "wb") as f:
@@ -95,7 +95,7 @@ def part_b():
     import multiprocessing
     cores = multiprocessing.cpu_count()
     ref = os.path.join(REPO, "oracle", "_ref", "indexsearcher")
-    for mc in (8192, 16384, 32768):
+    for mc in (8192, 16384):
         outb = os.path.join(idx, f"ref_mc{mc}.bin")
         r = subprocess.run([ref, "-d", str(qs.shape[1]), "-v", "Float",
                             "-f", "DEFAULT", "-i", qfile, "-x", idx,
