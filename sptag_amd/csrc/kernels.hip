@@ -553,7 +553,24 @@ DEV void search_trees_dev(QCtx<T>& c, int limit)
                 int cnt = min(64, ce - base);
                 if (c.lane < cnt) c.istage[c.lane] = di.tree_nodes[(size_t)(base + c.lane) * 3];
                 __syncthreads();
+                /* Issue ALL children's vector lines before the staged
+                 * distance rounds start: every one of them is needed (no
+                 * speculation waste), so the 16 round-pipeline latencies
+                 * collapse into one. Seed-phase tree-center gathers were
+                 * 31% of kernel cycles. */
+                uint32_t ts = 0;
+                if (c.lane < cnt) {
+                    const char* vp =
+                        (const char*)vec_at<T>(di, c.istage[c.lane]);
+                    const int nl =
+                        ((int)((unsigned)di.dim * sizeof(T)) + 127) >> 7;
+                    for (int l = 0; l < nl; l++)
+                        ts += (uint32_t)(uint8_t)vp[(size_t)l << 7];
+                }
                 stage_dists<T, DM>(c, cnt);
+                asm volatile("" :: "v"(ts));   /* sink after the staging so
+                                                  the touches and the first
+                                                  rounds overlap */
                 spt_warm_parents(c, cnt);
                 if (c.lane == 0)
                     for (int i = 0; i < cnt; i++)
@@ -588,7 +605,24 @@ DEV void init_search_trees_dev(QCtx<T>& c)
                 int cnt = min(64, ce - base);
                 if (c.lane < cnt) c.istage[c.lane] = di.tree_nodes[(size_t)(base + c.lane) * 3];
                 __syncthreads();
+                /* Issue ALL children's vector lines before the staged
+                 * distance rounds start: every one of them is needed (no
+                 * speculation waste), so the 16 round-pipeline latencies
+                 * collapse into one. Seed-phase tree-center gathers were
+                 * 31% of kernel cycles. */
+                uint32_t ts = 0;
+                if (c.lane < cnt) {
+                    const char* vp =
+                        (const char*)vec_at<T>(di, c.istage[c.lane]);
+                    const int nl =
+                        ((int)((unsigned)di.dim * sizeof(T)) + 127) >> 7;
+                    for (int l = 0; l < nl; l++)
+                        ts += (uint32_t)(uint8_t)vp[(size_t)l << 7];
+                }
                 stage_dists<T, DM>(c, cnt);
+                asm volatile("" :: "v"(ts));   /* sink after the staging so
+                                                  the touches and the first
+                                                  rounds overlap */
                 spt_warm_parents(c, cnt);
                 if (c.lane == 0)
                     for (int i = 0; i < cnt; i++)
