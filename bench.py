@@ -499,7 +499,7 @@ def main():
             "metric": "QPS @ recall@10>=0.95, batch=10k queries",
             "value": round(value, 1),
             "unit": "queries/s",
-            "n_gpus": world,
+            "n_gpus": 1 if forced_shard else world,
             "steps": args.steps,
             "warmup": args.warmup,
             "ms_per_step": round(ms_per_step, 3),
@@ -514,7 +514,10 @@ def main():
                 "nq": nq, "k": k, "max_check": chosen_mc,
                 "recall_at_10": round(chosen_recall, 4),
                 "recall_gate_met": bool(chosen_recall >= 0.95),
-                "sharding": "contiguous VID ranges" if world > 1 else "none",
+                "sharding": (f"forced shard {rank}/{world} (1 process, "
+                             "shard-local)" if forced_shard else
+                             "contiguous VID ranges" if world > 1 else
+                             "none"),
             },
             "roofline": {
                 "bound": "hbm",
