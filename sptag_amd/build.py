@@ -534,20 +534,32 @@ def build_rng_graph(vectors, *, degree=32, ntrees=4, tpt_leaf=1000, cand=256,
     bknn = min(8, L - 1)
     bridge_ids = None
     if bknn > 0:
-        leaf_of = torch.zeros(n, dtype=torch.int64, device=device)
-        maxm = max(c for _, c in bounds)
-        pad = torch.zeros((L, maxm), dtype=torch.int64, device=device)
-        pmask = torch.zeros((L, maxm), dtype=torch.bool, device=device)
-        for li, (st, c) in enumerate(bounds):
-            leaf_of[perm[st:st + c]] = li
-            pad[li, :c] = perm[st:st + c]
-            pmask[li, :c] = True
-        pts = xf[pad]                                    # [L, maxm, d]
-        cnt = pmask.sum(1, keepdim=True).clamp(min=1)
-        centroids = (pts * pmask[:, :, None]).sum(1) / cnt
-        cd = ((pts - centroids[:, None, :]) ** 2).sum(-1)
-        cd = cd.masked_fill(~pmask, float("inf"))
-        medoid = pad.gather(1, cd.argmin(1, keepdim=True)).squeeze(1)  # [L]
+        # Leaf centroids + medoids WITHOUT padded [L, maxleaf, d] tensors
+        # (at 100M/131k leaves those transients alone are ~90 GB and OOM
+        # next to the candidate pools): index_add centroids, then a
+        # packed-key (float-bits<<32 | index) segmented argmin per leaf.
+        d = xf.shape[1]
+        counts_t = torch.tensor([c for _, c in bounds], device=device)
+        leaf_of_perm = torch.repeat_interleave(
+            torch.arange(L, device=device), counts_t)
+        leaf_of = torch.empty(n, dtype=torch.int64, device=device)
+        leaf_of[perm] = leaf_of_perm
+        centroids = torch.zeros((L, d), device=device)
+        pc = 10_000_000
+        for s0 in range(0, n, pc):
+            e0 = min(n, s0 + pc)
+            centroids.index_add_(0, leaf_of_perm[s0:e0], xf[perm[s0:e0]])
+        centroids /= counts_t.float().clamp(min=1)[:, None]
+        keymin = torch.full((L,), 2 ** 62, dtype=torch.int64, device=device)
+        for s0 in range(0, n, pc):
+            e0 = min(n, s0 + pc)
+            lf = leaf_of_perm[s0:e0]
+            dc = ((xf[perm[s0:e0]] - centroids[lf]) ** 2).sum(1)
+            # non-negative f32 bit patterns are order-preserving as ints
+            key = (dc.view(torch.int32).to(torch.int64) << 32) | \
+                torch.arange(s0, e0, device=device)
+            keymin.scatter_reduce_(0, lf, key, reduce="amin")
+        medoid = perm[keymin & 0xffffffff]               # [L]
         # exponential-rank ladder of nearest leaves: ranks 1,2,4,...: short
         # bridges link adjacent leaves, long ones cross cluster groups —
         # scale-independent (at 100M a tight cluster spans many leaves, so
@@ -564,7 +576,7 @@ def build_rng_graph(vectors, *, degree=32, ntrees=4, tpt_leaf=1000, cand=256,
             # drop self (rank 0), take the ladder ranks
             nleaf[s0:e0] = idxs[:, 1:][:, [r - 1 for r in ranks]]
         bridge_ids = medoid[nleaf][leaf_of].int()             # [n, len(ranks)]
-        del pad, pmask, pts, cd
+        del leaf_of_perm, keymin, centroids
 
     if point_chunk is None:
         point_chunk = _prune_chunk(cand, xf.shape[1])
