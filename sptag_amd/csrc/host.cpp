@@ -430,7 +430,9 @@ static int search_device_core(SptagAmdIndex* ix, const void* d_q, int32_t nq,
      * saturate. Off by default; kept for experiments. */
     cfg.spec_flags = 0;
     if (const char* e = getenv("SPTAG_AMD_SPEC")) cfg.spec_flags = atoi(e);
-    cfg.prof = getenv("SPTAG_AMD_PROF") ? 1 : 0;   /* phase-cycle diagnostic */
+    /* phase-cycle diagnostic (BKT kernels only — the KDT kernel writes
+     * plain stride-2 stats) */
+    cfg.prof = (getenv("SPTAG_AMD_PROF") && ix->algo == ALGO_BKT) ? 1 : 0;
     const int sstride = cfg.prof ? PROF_STATS : 2;
 
     int lds_limit = 64 * 1024;
