@@ -22,6 +22,7 @@
 #include <cstdio>
 #include <cstdlib>
 #include <algorithm>
+#include <cctype>
 #include <cstring>
 #include <mutex>
 #include <string>
@@ -106,26 +107,52 @@ bool read_file(const std::string& path, std::vector<char>& out)
     return ok;
 }
 
+/* SimpleIniReader-compatible lookup (reference src/Helper/
+ * SimpleIniReader.cpp): case-insensitive section and key names, leading/
+ * trailing whitespace around the key and value tolerated, ';' comment
+ * lines skipped. This is the drop-in seam, so it accepts what the
+ * reference's own parser would. */
+static bool ieq(const std::string& a, const std::string& b)
+{
+    if (a.size() != b.size()) return false;
+    for (size_t i = 0; i < a.size(); i++)
+        if (tolower((unsigned char)a[i]) != tolower((unsigned char)b[i]))
+            return false;
+    return true;
+}
+
+static std::string trim(const std::string& s)
+{
+    size_t b = s.find_first_not_of(" \t\r");
+    if (b == std::string::npos) return "";
+    size_t e = s.find_last_not_of(" \t\r");
+    return s.substr(b, e - b + 1);
+}
+
 bool ini_get(const std::string& text, const char* section, const char* key,
              std::string& out)
 {
-    std::string sect = std::string("[") + section + "]";
-    size_t p = text.find(sect);
-    if (p == std::string::npos) return false;
-    p += sect.size();
-    size_t end = text.find('[', p);
-    std::string k = std::string(key) + "=";
-    while (p < text.size() && (end == std::string::npos || p < end)) {
+    bool in_section = false;
+    size_t p = 0;
+    while (p < text.size()) {
         size_t eol = text.find('\n', p);
         if (eol == std::string::npos) eol = text.size();
-        while (p < eol && (text[p] == ' ' || text[p] == '\r')) p++;
-        if (text.compare(p, k.size(), k) == 0) {
-            out = text.substr(p + k.size(), eol - p - k.size());
-            while (!out.empty() && (out.back() == '\r' || out.back() == ' '))
-                out.pop_back();
+        std::string line = trim(text.substr(p, eol - p));
+        p = eol + 1;
+        if (line.empty() || line[0] == ';') continue;
+        if (line.front() == '[') {
+            size_t close = line.find(']');
+            if (close == std::string::npos) continue;
+            in_section = ieq(trim(line.substr(1, close - 1)), section);
+            continue;
+        }
+        if (!in_section) continue;
+        size_t eq = line.find('=');
+        if (eq == std::string::npos) continue;
+        if (ieq(trim(line.substr(0, eq)), key)) {
+            out = trim(line.substr(eq + 1));
             return true;
         }
-        p = eol + 1;
     }
     return false;
 }

@@ -178,3 +178,32 @@ def test_metadata_set(tmp_path):
     ix = AnnIndex.Load(str(idx2))
     assert ix.metadata is not None
     assert ix.metadata.get(2) == b"gamma-123"
+
+
+def test_load_tolerant_ini(tmp_path):
+    """The ini parser at the drop-in seam accepts what the reference's
+    SimpleIniReader accepts: case-insensitive section/key names, spaces
+    around '=', ';' comment lines (src/Helper/SimpleIniReader.cpp)."""
+    import shutil
+    from conftest import golden_fixtures, load_golden
+    names = [n for n in golden_fixtures() if n.startswith("f32_l2_n")]
+    if not names:
+        pytest.skip("no golden fixtures")
+    g = load_golden(names[0])
+    dst = tmp_path / "index"
+    shutil.copytree(g["index"], dst)
+    ini = (dst / "indexloader.ini").read_text()
+    # mangle: comments, case, whitespace — still reference-legal
+    mangled = ["; rewritten by test", "[index]"]
+    for line in ini.splitlines():
+        line = line.strip()
+        if not line or line.startswith("["):
+            continue
+        if "=" in line:
+            k, v = line.split("=", 1)
+            mangled.append(f"  {k.lower()} = {v}")
+    (dst / "indexloader.ini").write_text("\n".join(mangled) + "\n")
+    ix = sptag_amd.AnnIndex.Load(str(dst))
+    assert ix.n == g["meta"]["n"]
+    assert ix.distmethod == sptag_amd.DM_L2
+    assert ix.default_maxcheck == 8192
